@@ -1,0 +1,34 @@
+"""pytorch_ps_mpi_amd — MI355X-native async parameter-server training engine.
+
+Brand-new implementation of the capabilities of stsievert/pytorch_ps_mpi
+(reference surveyed in SURVEY.md): a drop-in torch.optim.Optimizer with a
+pluggable gradient-compression codec, replicated / sync-PS / AsySG-InCon
+exchange modes — re-architected for one 8×MI355X node: flat device buffers,
+RCCL over xGMI (torch.distributed "nccl"), hand-written CDNA4 HIP kernels
+for every hot op (ops/csrc/ps_kernels.hip), fp32 master weights under bf16
+compute.
+
+Public API (reference parity: __init__.py:1 exported MPI_PS, Adam, SGD):
+
+    from pytorch_ps_mpi_amd import PS, SGD, Adam
+    opt = SGD(model.named_parameters(), lr=0.1, momentum=0.9,
+              code="topk:0.01", mode="async")
+    loss.backward(); loss, metrics = opt.step(loss=loss)
+"""
+
+from . import codecs, models, ops
+from .codecs import Identity, QuantInt8, TopK, get_codec
+from .optim import PS, SGD, Adam
+from .parallel.comm import Comm, init_distributed
+
+# reference-compatible alias (ps.py:53 class MPI_PS)
+MPI_PS = PS
+
+__all__ = [
+    "PS", "MPI_PS", "SGD", "Adam",
+    "Identity", "TopK", "QuantInt8", "get_codec",
+    "Comm", "init_distributed",
+    "codecs", "models", "ops",
+]
+
+__version__ = "0.1.0"

@@ -1,0 +1,171 @@
+"""Gradient codecs — the device-resident successor of the reference's
+`codings` plugin contract (ps.py:18).
+
+A codec turns a flat gradient span (model dtype) into a FIXED-CAPACITY wire
+tensor and back.  Fixed capacity is what makes RCCL practical: there is no
+gatherv over xGMI, so every rank's message for a bucket has identical size
+(the reference instead ran a per-parameter size all-gather + Iallgatherv,
+mpi_comms.py:150-163 — pure latency it did not need).
+
+Contract (all tensors device-resident, no host round trips):
+    wire_dtype                      -> torch dtype of the wire tensor
+    wire_numel(bucket_numel)        -> fixed wire length for a bucket
+    encode(src, wire)               -> fill wire from src (model-dtype flat)
+    decode_reduce(dst, wires, gscale, beta)
+        dst(f32) = beta*dst + gscale * sum_r decode(wires[r])
+        summed in rank order -> bitwise deterministic on every rank.
+    bytes_on_wire(bucket_numel)     -> payload bytes (metrics)
+
+Codecs are stateless w.r.t. step; per-device scratch is cached internally.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from . import ops
+
+
+class Identity:
+    """No compression: wire is the raw model-dtype gradient."""
+
+    name = "identity"
+    supports_allreduce = True
+
+    def wire_dtype(self, src_dtype):
+        return src_dtype
+
+    def wire_numel(self, numel):
+        return numel
+
+    def encode(self, src, wire):
+        if wire.data_ptr() != src.data_ptr():
+            wire.copy_(src)
+
+    def decode_reduce(self, dst, wires, gscale=1.0, beta=0.0, src_dtype=None):
+        ops.reduce_accum(dst, list(wires), scale=gscale, beta=beta)
+
+    def bytes_on_wire(self, numel, dtype=torch.bfloat16):
+        return numel * dtype.itemsize
+
+
+class TopK:
+    """Magnitude top-k sparsification (density fraction per bucket).
+
+    Wire layout (uint8): [ int32 idx[k] | val[k] (model dtype) | pad ].
+    k is fixed per bucket size -> fixed wire capacity.
+    """
+
+    name = "topk"
+    supports_allreduce = False
+
+    def __init__(self, density=0.01, min_k=8):
+        if not (0.0 < density <= 1.0):
+            raise ValueError("density in (0,1]")
+        self.density = density
+        self.min_k = min_k
+        self._ws = {}
+
+    def k_for(self, numel):
+        k = max(self.min_k, int(numel * self.density))
+        return min(k, numel)
+
+    def wire_dtype(self, src_dtype):
+        return torch.uint8
+
+    def _esize(self, dtype):
+        return dtype.itemsize
+
+    def wire_numel(self, numel, src_dtype=torch.bfloat16):
+        k = self.k_for(numel)
+        raw = 4 * k + self._esize(src_dtype) * k
+        return (raw + 15) // 16 * 16
+
+    def _views(self, wire, numel, src_dtype):
+        k = self.k_for(numel)
+        idx = wire[:4 * k].view(torch.int32)
+        vbytes = self._esize(src_dtype) * k
+        val = wire[4 * k:4 * k + vbytes].view(src_dtype)
+        return k, idx, val
+
+    def _workspace(self, device):
+        key = str(device)
+        if key not in self._ws:
+            self._ws[key] = ops.topk_workspace(device)
+        return self._ws[key]
+
+    def encode(self, src, wire):
+        k, idx, val = self._views(wire, src.numel(), src.dtype)
+        ops.topk_encode(src, k, self._workspace(src.device), idx, val)
+
+    def decode_reduce(self, dst, wires, gscale=1.0, beta=0.0, src_dtype=None):
+        numel = dst.numel()
+        dt = src_dtype if src_dtype is not None else (
+            torch.bfloat16 if dst.is_cuda else torch.float32)
+        if beta == 0.0:
+            dst.zero_()
+        elif beta != 1.0:
+            dst.mul_(beta)
+        for w in wires:  # one message at a time: unique indices, deterministic
+            k, idx, val = self._views(w, numel, dt)
+            ops.topk_scatter(dst, idx, val, k, gscale)
+
+    def bytes_on_wire(self, numel, dtype=torch.bfloat16):
+        return self.wire_numel(numel, dtype)
+
+
+class QuantInt8:
+    """Per-256-element-chunk absmax int8 quantization.
+
+    Wire layout (uint8): [ f32 scales[nchunks] | int8 q[numel] | pad ].
+    """
+
+    name = "quant8"
+    supports_allreduce = False
+
+    def wire_dtype(self, src_dtype):
+        return torch.uint8
+
+    def wire_numel(self, numel, src_dtype=None):
+        nc = ops.quant8_nscales(numel)
+        raw = 4 * nc + numel
+        return (raw + 15) // 16 * 16
+
+    def _views(self, wire, numel):
+        nc = ops.quant8_nscales(numel)
+        scales = wire[:4 * nc].view(torch.float32)
+        q = wire[4 * nc:4 * nc + numel].view(torch.int8)
+        return scales, q
+
+    def encode(self, src, wire):
+        scales, q = self._views(wire, src.numel())
+        ops.quant8_encode(src, scales, q)
+
+    def decode_reduce(self, dst, wires, gscale=1.0, beta=0.0, src_dtype=None):
+        numel = dst.numel()
+        sc, qs = [], []
+        for w in wires:
+            s, q = self._views(w, numel)
+            sc.append(s)
+            qs.append(q)
+        ops.quant8_reduce(dst, sc, qs, gscale=gscale, beta=beta)
+
+    def bytes_on_wire(self, numel, dtype=None):
+        return self.wire_numel(numel)
+
+
+def get_codec(spec):
+    """'identity' | 'topk' | 'topk:0.02' | 'quant8' | codec instance | None."""
+    if spec is None:
+        return Identity()
+    if not isinstance(spec, str):
+        return spec
+    if spec == "identity":
+        return Identity()
+    if spec.startswith("topk"):
+        parts = spec.split(":")
+        density = float(parts[1]) if len(parts) > 1 else 0.01
+        return TopK(density=density)
+    if spec in ("quant8", "int8"):
+        return QuantInt8()
+    raise ValueError(f"unknown codec {spec!r}")

@@ -1,0 +1,69 @@
+"""GPT-2-small (Radford et al. 2019) for BASELINE config 5
+(AsySG-InCon stale-gradient path)."""
+
+from __future__ import annotations
+
+import math
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+class CausalBlock(nn.Module):
+    def __init__(self, dim, heads):
+        super().__init__()
+        self.n1 = nn.LayerNorm(dim)
+        self.qkv = nn.Linear(dim, 3 * dim)
+        self.proj = nn.Linear(dim, dim)
+        self.n2 = nn.LayerNorm(dim)
+        self.fc1 = nn.Linear(dim, 4 * dim)
+        self.fc2 = nn.Linear(4 * dim, dim)
+        self.heads = heads
+
+    def forward(self, x):
+        B, T, D = x.shape
+        qkv = self.qkv(self.n1(x)).view(B, T, 3, self.heads, D // self.heads)
+        q, k, v = qkv.permute(2, 0, 3, 1, 4)
+        y = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+        y = y.transpose(1, 2).reshape(B, T, D)
+        x = x + self.proj(y)
+        x = x + self.fc2(F.gelu(self.fc1(self.n2(x))))
+        return x
+
+
+class GPT2(nn.Module):
+    def __init__(self, vocab=50257, ctx=1024, dim=768, depth=12, heads=12):
+        super().__init__()
+        self.wte = nn.Embedding(vocab, dim)
+        self.wpe = nn.Embedding(ctx, dim)
+        self.blocks = nn.ModuleList(CausalBlock(dim, heads)
+                                    for _ in range(depth))
+        self.norm = nn.LayerNorm(dim)
+        self.ctx = ctx
+        for m in self.modules():
+            if isinstance(m, nn.Linear):
+                nn.init.normal_(m.weight, std=0.02)
+                if m.bias is not None:
+                    nn.init.zeros_(m.bias)
+            elif isinstance(m, nn.Embedding):
+                nn.init.normal_(m.weight, std=0.02)
+
+    def forward(self, idx):
+        B, T = idx.shape
+        pos = torch.arange(T, device=idx.device)
+        x = self.wte(idx) + self.wpe(pos)[None]
+        for blk in self.blocks:
+            x = blk(x)
+        x = self.norm(x)
+        # tied output head
+        return x @ self.wte.weight.t()
+
+    def loss(self, idx, targets):
+        logits = self.forward(idx)
+        return F.cross_entropy(logits.view(-1, logits.size(-1)).float(),
+                               targets.reshape(-1))
+
+
+def gpt2_small(vocab=50257, ctx=1024):
+    return GPT2(vocab=vocab, ctx=ctx)

@@ -1,0 +1,55 @@
+"""ViT-B/16 (Dosovitskiy et al. 2020) for BASELINE config 4."""
+
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+class Block(nn.Module):
+    def __init__(self, dim, heads, mlp_ratio=4.0):
+        super().__init__()
+        self.n1 = nn.LayerNorm(dim)
+        self.qkv = nn.Linear(dim, dim * 3)
+        self.proj = nn.Linear(dim, dim)
+        self.heads = heads
+        self.n2 = nn.LayerNorm(dim)
+        h = int(dim * mlp_ratio)
+        self.fc1 = nn.Linear(dim, h)
+        self.fc2 = nn.Linear(h, dim)
+
+    def forward(self, x):
+        B, N, D = x.shape
+        qkv = self.qkv(self.n1(x)).view(B, N, 3, self.heads, D // self.heads)
+        q, k, v = qkv.permute(2, 0, 3, 1, 4)
+        y = F.scaled_dot_product_attention(q, k, v)
+        y = y.transpose(1, 2).reshape(B, N, D)
+        x = x + self.proj(y)
+        x = x + self.fc2(F.gelu(self.fc1(self.n2(x))))
+        return x
+
+
+class ViT(nn.Module):
+    def __init__(self, img=224, patch=16, dim=768, depth=12, heads=12,
+                 num_classes=1000):
+        super().__init__()
+        self.patch_embed = nn.Conv2d(3, dim, patch, patch)
+        n = (img // patch) ** 2
+        self.cls = nn.Parameter(torch.zeros(1, 1, dim))
+        self.pos = nn.Parameter(torch.randn(1, n + 1, dim) * 0.02)
+        self.blocks = nn.ModuleList(Block(dim, heads) for _ in range(depth))
+        self.norm = nn.LayerNorm(dim)
+        self.head = nn.Linear(dim, num_classes)
+
+    def forward(self, x):
+        x = self.patch_embed(x).flatten(2).transpose(1, 2)
+        cls = self.cls.expand(x.shape[0], -1, -1)
+        x = torch.cat([cls, x], dim=1) + self.pos
+        for blk in self.blocks:
+            x = blk(x)
+        return self.head(self.norm(x)[:, 0])
+
+
+def vit_b16(num_classes=1000):
+    return ViT(dim=768, depth=12, heads=12, num_classes=num_classes)

@@ -1,0 +1,193 @@
+"""Device-dispatching op layer.
+
+On GPU (ROCm) every function calls the in-tree gfx950 HIP extension
+(`_ps_hip.so`) and FAILS LOUDLY if it is missing — there is no silent eager
+fallback on a GPU box.  On CPU the same semantics are provided by plain torch
+reference implementations (used by the gloo multi-process tests and as the
+numerics oracle for the HIP kernels).
+
+These ops replace the reference's Python hot path
+(stsievert/pytorch_ps_mpi): cross-rank grad sum ps.py:176, SGD math
+ps.py:197-214, Adam math ps.py:218-261, and the external `codings`
+compression plugin (ps.py:18).
+"""
+
+from __future__ import annotations
+
+import torch
+
+_EXT = None
+_EXT_ERR = None
+try:
+    from . import _ps_hip as _EXT  # type: ignore
+except Exception as e:  # pragma: no cover - exercised only without built ext
+    _EXT_ERR = e
+
+HAVE_EXT = _EXT is not None
+
+QCHUNK = 256
+TOPK_WS_WORDS = 2048 + 3 + 2
+
+
+def _require_ext(t: torch.Tensor):
+    if t.is_cuda:
+        if not HAVE_EXT:
+            raise RuntimeError(
+                "pytorch_ps_mpi_amd HIP extension (_ps_hip) is not built but a "
+                "GPU tensor reached the op layer. Build it with "
+                "`PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace`. "
+                f"Import error was: {_EXT_ERR!r}"
+            )
+        return True
+    return False
+
+
+# ---------------------------------------------------------------------------
+# fused optimizers
+# ---------------------------------------------------------------------------
+
+def fused_sgd(p, buf, g, p_out, *, lr, momentum=0.0, dampening=0.0, wd=0.0,
+              nesterov=False, mom_init=False, gscale=1.0):
+    """p/buf/g are flat fp32; p_out optional bf16/f32 model copy."""
+    if _require_ext(p):
+        _EXT.fused_sgd(p, buf, g, p_out, lr, momentum, dampening, wd,
+                       bool(nesterov), bool(mom_init), gscale)
+        return
+    d = g * gscale
+    if wd != 0.0:
+        d = d.add(p, alpha=wd)
+    if momentum != 0.0:
+        if mom_init:
+            buf.copy_(d)
+        else:
+            buf.mul_(momentum).add_(d, alpha=1.0 - dampening)
+        d = d.add(buf, alpha=momentum) if nesterov else buf
+    p.add_(d, alpha=-lr)
+    if p_out is not None:
+        p_out.copy_(p)
+
+
+def fused_adam(p, m1, m2, vmax, g, p_out, *, lr, beta1=0.9, beta2=0.999,
+               eps=1e-8, wd=0.0, step=1, amsgrad=False, gscale=1.0):
+    if _require_ext(p):
+        _EXT.fused_adam(p, m1, m2, vmax, g, p_out, lr, beta1, beta2, eps, wd,
+                        int(step), bool(amsgrad), gscale)
+        return
+    d = g * gscale
+    if wd != 0.0:
+        d = d.add(p, alpha=wd)
+    m1.mul_(beta1).add_(d, alpha=1.0 - beta1)
+    m2.mul_(beta2).addcmul_(d, d, value=1.0 - beta2)
+    bc1 = 1.0 - beta1 ** step
+    bc2 = 1.0 - beta2 ** step
+    v = m2
+    if amsgrad:
+        torch.maximum(vmax, m2, out=vmax)
+        v = vmax
+    denom = (v.sqrt() / (bc2 ** 0.5)).add_(eps)
+    p.addcdiv_(m1, denom, value=-(lr / bc1))
+    if p_out is not None:
+        p_out.copy_(p)
+
+
+# ---------------------------------------------------------------------------
+# dense multi-source reduction (deterministic source order)
+# ---------------------------------------------------------------------------
+
+def reduce_accum(dst, srcs, scale=1.0, beta=0.0):
+    """dst(f32) = beta*dst + scale * sum_r srcs[r]  (srcs all bf16 or all f32)."""
+    if _require_ext(dst):
+        _EXT.reduce_accum(dst, list(srcs), scale, beta)
+        return
+    acc = torch.zeros_like(dst)
+    for s in srcs:
+        acc += s.float()
+    if beta == 0.0:
+        dst.copy_(acc * scale)
+    else:
+        dst.mul_(beta).add_(acc, alpha=scale)
+
+
+def f32_to_bf16(src, dst):
+    if _require_ext(src):
+        _EXT.f32_to_bf16(src, dst)
+        return
+    dst.copy_(src)
+
+
+def bf16_to_f32(src, dst):
+    if _require_ext(dst):
+        _EXT.bf16_to_f32(src, dst)
+        return
+    dst.copy_(src)
+
+
+# ---------------------------------------------------------------------------
+# int8 gradient quantization (per-256-element absmax chunks)
+# ---------------------------------------------------------------------------
+
+def quant8_nscales(n: int) -> int:
+    return (n + QCHUNK - 1) // QCHUNK
+
+
+def quant8_encode(src, scales, q):
+    if _require_ext(src):
+        _EXT.quant8_encode(src, scales, q)
+        return
+    n = src.numel()
+    nc = quant8_nscales(n)
+    pad = nc * QCHUNK - n
+    x = src.float()
+    if pad:
+        x = torch.cat([x, x.new_zeros(pad)])
+    x = x.view(nc, QCHUNK)
+    absmax = x.abs().amax(dim=1)
+    s = torch.where(absmax > 0, absmax / 127.0, torch.ones_like(absmax))
+    scales[:nc].copy_(s)
+    qv = torch.clamp(torch.round(x / s[:, None]), -127, 127).to(torch.int8)
+    q[:n].copy_(qv.view(-1)[:n])
+
+
+def quant8_reduce(dst, scales_list, qs_list, gscale=1.0, beta=0.0):
+    if _require_ext(dst):
+        _EXT.quant8_reduce(dst, list(scales_list), list(qs_list), gscale, beta)
+        return
+    n = dst.numel()
+    acc = torch.zeros_like(dst)
+    for s, q in zip(scales_list, qs_list):
+        sc = s.repeat_interleave(QCHUNK)[:n]
+        acc += sc * q[:n].float()
+    if beta == 0.0:
+        dst.copy_(acc * gscale)
+    else:
+        dst.mul_(beta).add_(acc, alpha=gscale)
+
+
+# ---------------------------------------------------------------------------
+# top-k magnitude sparsification
+# ---------------------------------------------------------------------------
+
+def topk_workspace(device) -> torch.Tensor:
+    return torch.zeros(TOPK_WS_WORDS, dtype=torch.int32, device=device)
+
+
+def topk_encode(src, k, ws, out_idx, out_val):
+    """Select k largest-|x| elements -> (out_idx int32, out_val src.dtype).
+
+    GPU: radix-style bin threshold (ties within the boundary bin are taken in
+    arbitrary order).  CPU: exact torch.topk.
+    """
+    if _require_ext(src):
+        _EXT.topk_encode(src, int(k), ws, out_idx, out_val)
+        return
+    v, idx = torch.topk(src.float().abs(), int(k), sorted=False)
+    out_idx[:k].copy_(idx.to(torch.int32))
+    out_val[:k].copy_(src[idx])
+
+
+def topk_scatter(dst, idx, val, k, gscale=1.0):
+    """dst[idx[:k]] += gscale * val[:k] — one message (unique indices)."""
+    if _require_ext(dst):
+        _EXT.topk_scatter(dst, idx, val, int(k), gscale)
+        return
+    dst.index_add_(0, idx[:k].long(), val[:k].float() * gscale)

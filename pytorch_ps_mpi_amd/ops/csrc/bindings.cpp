@@ -1,0 +1,238 @@
+// bindings.cpp — torch extension surface for the CDNA4 PS kernels.
+// Pure HIP-side code (no CUDA compat paths): streams come from c10::hip.
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#include <cstdint>
+#include <vector>
+
+extern "C" {
+int ps_fused_sgd(void* stream, float* p, float* buf, const float* g, void* p_out,
+                 int p_out_is_bf16, int64_t n, float lr, float momentum,
+                 float dampening, float wd, int nesterov, int mom_init, float gscale);
+int ps_fused_adam(void* stream, float* p, float* m1, float* m2, float* vmax,
+                  const float* g, void* p_out, int p_out_is_bf16, int64_t n,
+                  float lr, float beta1, float beta2, float eps, float wd,
+                  float bc1, float bc2_sqrt, int amsgrad, float gscale);
+int ps_reduce_accum(void* stream, float* dst, const void** srcs, int nsrc,
+                    int src_is_bf16, int64_t n, float scale, float beta);
+int ps_f32_to_bf16(void* stream, const float* src, void* dst, int64_t n);
+int ps_bf16_to_f32(void* stream, const void* src, float* dst, int64_t n);
+int ps_quant8_encode(void* stream, const void* src, int src_is_bf16,
+                     float* scales, int8_t* q, int64_t n);
+int ps_quant8_reduce(void* stream, float* dst, const void** scales,
+                     const void** qs, int nsrc, int64_t n, float gscale, float beta);
+int ps_topk_workspace_words(void);
+int ps_topk_encode(void* stream, const void* src, int src_is_bf16, int64_t n,
+                   int64_t k, uint32_t* ws, int32_t* out_idx, void* out_val);
+int ps_topk_scatter(void* stream, float* dst, const int32_t* idx,
+                    const void* val, int val_is_bf16, int64_t k, float gscale);
+}
+
+namespace {
+
+void* cur_stream(const at::Tensor& t) {
+  return (void*)c10::hip::getCurrentHIPStream(t.get_device()).stream();
+}
+
+void check_flat(const at::Tensor& t, at::ScalarType dt, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == dt, name, " has wrong dtype");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+int is_bf16(const at::Tensor& t) { return t.scalar_type() == at::kBFloat16 ? 1 : 0; }
+
+void throw_on(int err, const char* what) {
+  TORCH_CHECK(err == 0, what, ": hip error ", err);
+}
+
+void fused_sgd(at::Tensor p, c10::optional<at::Tensor> buf, at::Tensor g,
+               c10::optional<at::Tensor> p_out, double lr, double momentum,
+               double dampening, double wd, bool nesterov, bool mom_init,
+               double gscale) {
+  check_flat(p, at::kFloat, "param");
+  check_flat(g, at::kFloat, "grad");
+  TORCH_CHECK(p.numel() == g.numel(), "param/grad length mismatch");
+  float* buf_ptr = nullptr;
+  if (momentum != 0.0) {
+    TORCH_CHECK(buf.has_value(), "momentum buffer required when momentum != 0");
+    check_flat(*buf, at::kFloat, "momentum buffer");
+    TORCH_CHECK(buf->numel() == p.numel(), "momentum buffer length mismatch");
+    buf_ptr = buf->data_ptr<float>();
+  }
+  void* out_ptr = nullptr;
+  int out_bf16 = 0;
+  if (p_out.has_value()) {
+    TORCH_CHECK(p_out->is_cuda() && p_out->is_contiguous(), "p_out invalid");
+    TORCH_CHECK(p_out->numel() == p.numel(), "p_out length mismatch");
+    out_ptr = p_out->data_ptr();
+    out_bf16 = is_bf16(*p_out);
+    TORCH_CHECK(out_bf16 || p_out->scalar_type() == at::kFloat, "p_out dtype");
+  }
+  throw_on(ps_fused_sgd(cur_stream(p), p.data_ptr<float>(), buf_ptr,
+                        g.data_ptr<float>(), out_ptr, out_bf16, p.numel(),
+                        (float)lr, (float)momentum, (float)dampening, (float)wd,
+                        nesterov ? 1 : 0, mom_init ? 1 : 0, (float)gscale),
+           "fused_sgd");
+}
+
+void fused_adam(at::Tensor p, at::Tensor m1, at::Tensor m2,
+                c10::optional<at::Tensor> vmax, at::Tensor g,
+                c10::optional<at::Tensor> p_out, double lr, double beta1,
+                double beta2, double eps, double wd, int64_t step, bool amsgrad,
+                double gscale) {
+  check_flat(p, at::kFloat, "param");
+  check_flat(m1, at::kFloat, "exp_avg");
+  check_flat(m2, at::kFloat, "exp_avg_sq");
+  check_flat(g, at::kFloat, "grad");
+  float* vmax_ptr = nullptr;
+  if (amsgrad) {
+    TORCH_CHECK(vmax.has_value(), "amsgrad requires max_exp_avg_sq");
+    check_flat(*vmax, at::kFloat, "max_exp_avg_sq");
+    vmax_ptr = vmax->data_ptr<float>();
+  }
+  void* out_ptr = nullptr;
+  int out_bf16 = 0;
+  if (p_out.has_value()) {
+    TORCH_CHECK(p_out->is_cuda() && p_out->is_contiguous(), "p_out invalid");
+    out_ptr = p_out->data_ptr();
+    out_bf16 = is_bf16(*p_out);
+  }
+  const double bc1 = 1.0 - std::pow(beta1, (double)step);
+  const double bc2_sqrt = std::sqrt(1.0 - std::pow(beta2, (double)step));
+  throw_on(ps_fused_adam(cur_stream(p), p.data_ptr<float>(), m1.data_ptr<float>(),
+                         m2.data_ptr<float>(), vmax_ptr, g.data_ptr<float>(),
+                         out_ptr, out_bf16, p.numel(), (float)lr, (float)beta1,
+                         (float)beta2, (float)eps, (float)wd, (float)bc1,
+                         (float)bc2_sqrt, amsgrad ? 1 : 0, (float)gscale),
+           "fused_adam");
+}
+
+void reduce_accum(at::Tensor dst, std::vector<at::Tensor> srcs, double scale,
+                  double beta) {
+  check_flat(dst, at::kFloat, "dst");
+  TORCH_CHECK(!srcs.empty() && srcs.size() <= 8, "1..8 sources");
+  const void* ptrs[8];
+  int bf = is_bf16(srcs[0]);
+  for (size_t i = 0; i < srcs.size(); ++i) {
+    TORCH_CHECK(srcs[i].is_cuda() && srcs[i].is_contiguous(), "src invalid");
+    TORCH_CHECK(srcs[i].numel() == dst.numel(), "src length mismatch");
+    TORCH_CHECK(is_bf16(srcs[i]) == bf &&
+                    (bf || srcs[i].scalar_type() == at::kFloat),
+                "all sources must share dtype (f32 or bf16)");
+    ptrs[i] = srcs[i].data_ptr();
+  }
+  throw_on(ps_reduce_accum(cur_stream(dst), dst.data_ptr<float>(), ptrs,
+                           (int)srcs.size(), bf, dst.numel(), (float)scale,
+                           (float)beta),
+           "reduce_accum");
+}
+
+void f32_to_bf16(at::Tensor src, at::Tensor dst) {
+  check_flat(src, at::kFloat, "src");
+  check_flat(dst, at::kBFloat16, "dst");
+  TORCH_CHECK(src.numel() == dst.numel(), "length mismatch");
+  throw_on(ps_f32_to_bf16(cur_stream(src), src.data_ptr<float>(), dst.data_ptr(),
+                          src.numel()),
+           "f32_to_bf16");
+}
+
+void bf16_to_f32(at::Tensor src, at::Tensor dst) {
+  check_flat(src, at::kBFloat16, "src");
+  check_flat(dst, at::kFloat, "dst");
+  TORCH_CHECK(src.numel() == dst.numel(), "length mismatch");
+  throw_on(ps_bf16_to_f32(cur_stream(src), src.data_ptr(), dst.data_ptr<float>(),
+                          src.numel()),
+           "bf16_to_f32");
+}
+
+void quant8_encode(at::Tensor src, at::Tensor scales, at::Tensor q) {
+  TORCH_CHECK(src.is_cuda() && src.is_contiguous(), "src invalid");
+  int bf = is_bf16(src);
+  TORCH_CHECK(bf || src.scalar_type() == at::kFloat, "src dtype");
+  check_flat(scales, at::kFloat, "scales");
+  TORCH_CHECK(q.scalar_type() == at::kChar && q.is_contiguous() && q.is_cuda(),
+              "q must be int8");
+  const int64_t n = src.numel();
+  TORCH_CHECK(scales.numel() >= (n + 255) / 256, "scales too small");
+  TORCH_CHECK(q.numel() >= n, "q too small");
+  throw_on(ps_quant8_encode(cur_stream(src), src.data_ptr(), bf,
+                            scales.data_ptr<float>(),
+                            (int8_t*)q.data_ptr(), n),
+           "quant8_encode");
+}
+
+void quant8_reduce(at::Tensor dst, std::vector<at::Tensor> scales,
+                   std::vector<at::Tensor> qs, double gscale, double beta) {
+  check_flat(dst, at::kFloat, "dst");
+  TORCH_CHECK(scales.size() == qs.size() && !qs.empty() && qs.size() <= 8,
+              "1..8 sources");
+  const void* sp[8];
+  const void* qp[8];
+  for (size_t i = 0; i < qs.size(); ++i) {
+    TORCH_CHECK(scales[i].is_cuda() && scales[i].is_contiguous() &&
+                    scales[i].scalar_type() == at::kFloat,
+                "scales invalid");
+    TORCH_CHECK(qs[i].is_cuda() && qs[i].is_contiguous() &&
+                    qs[i].scalar_type() == at::kChar,
+                "q invalid");
+    TORCH_CHECK(qs[i].numel() >= dst.numel(), "q length mismatch");
+    sp[i] = scales[i].data_ptr();
+    qp[i] = qs[i].data_ptr();
+  }
+  throw_on(ps_quant8_reduce(cur_stream(dst), dst.data_ptr<float>(), sp, qp,
+                            (int)qs.size(), dst.numel(), (float)gscale,
+                            (float)beta),
+           "quant8_reduce");
+}
+
+int64_t topk_workspace_words() { return ps_topk_workspace_words(); }
+
+void topk_encode(at::Tensor src, int64_t k, at::Tensor ws, at::Tensor out_idx,
+                 at::Tensor out_val) {
+  TORCH_CHECK(src.is_cuda() && src.is_contiguous(), "src invalid");
+  int bf = is_bf16(src);
+  TORCH_CHECK(bf || src.scalar_type() == at::kFloat, "src dtype");
+  TORCH_CHECK(ws.scalar_type() == at::kInt || ws.scalar_type() == at::kUInt32,
+              "ws must be int32/uint32");
+  TORCH_CHECK(ws.numel() >= ps_topk_workspace_words(), "ws too small");
+  TORCH_CHECK(out_idx.scalar_type() == at::kInt && out_idx.numel() >= k,
+              "out_idx invalid");
+  TORCH_CHECK(out_val.scalar_type() == src.scalar_type() && out_val.numel() >= k,
+              "out_val invalid");
+  TORCH_CHECK(k >= 1 && k <= src.numel(), "k out of range");
+  throw_on(ps_topk_encode(cur_stream(src), src.data_ptr(), bf, src.numel(), k,
+                          (uint32_t*)ws.data_ptr(),
+                          (int32_t*)out_idx.data_ptr(), out_val.data_ptr()),
+           "topk_encode");
+}
+
+void topk_scatter(at::Tensor dst, at::Tensor idx, at::Tensor val, int64_t k,
+                  double gscale) {
+  check_flat(dst, at::kFloat, "dst");
+  TORCH_CHECK(idx.scalar_type() == at::kInt && idx.is_cuda(), "idx invalid");
+  int bf = is_bf16(val);
+  TORCH_CHECK(bf || val.scalar_type() == at::kFloat, "val dtype");
+  TORCH_CHECK(idx.numel() >= k && val.numel() >= k, "k too large");
+  throw_on(ps_topk_scatter(cur_stream(dst), dst.data_ptr<float>(),
+                           (const int32_t*)idx.data_ptr(), val.data_ptr(), bf, k,
+                           (float)gscale),
+           "topk_scatter");
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("fused_sgd", &fused_sgd, "fused SGD over flat fp32 buffers");
+  m.def("fused_adam", &fused_adam, "fused Adam over flat fp32 buffers");
+  m.def("reduce_accum", &reduce_accum, "dst = beta*dst + scale*sum(srcs)");
+  m.def("f32_to_bf16", &f32_to_bf16, "flat cast");
+  m.def("bf16_to_f32", &bf16_to_f32, "flat cast");
+  m.def("quant8_encode", &quant8_encode, "per-256-chunk absmax int8 quantize");
+  m.def("quant8_reduce", &quant8_reduce, "dequant+sum int8 messages");
+  m.def("topk_workspace_words", &topk_workspace_words);
+  m.def("topk_encode", &topk_encode, "magnitude top-k -> (idx, val)");
+  m.def("topk_scatter", &topk_scatter, "dst[idx] += gscale*val (one message)");
+}

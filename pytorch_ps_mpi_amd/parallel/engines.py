@@ -1,0 +1,194 @@
+"""Synchronous exchange engines.
+
+Three modes mirroring the reference's two wire protocols plus
+the degenerate single-process case:
+
+  LocalEngine       world_size == 1 — no comm, grads -> fp32 agg -> update.
+  ReplicatedEngine  the reference's SHIPPED path (ps.py:140-190): every rank
+                    receives every rank's encoded grad, decodes, sums and
+                    applies the identical update -> params stay bitwise equal
+                    by determinism.  Identity codec takes the all-reduce fast
+                    path; lossy codecs use fixed-capacity all-gather.
+  SyncPSEngine      the reference's README plan (README.md:37-46): gather
+                    encoded grads to the PS (rank 0), PS decodes+sums+applies,
+                    then broadcasts updated params.
+
+All buffers are flat device spans; comms are issued per bucket in fixed
+bucket order on every rank (the reference instead barriered on all encodes —
+ps.py:129 — and serialized the exchange).
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.distributed as dist
+
+from .. import ops
+
+
+class LocalEngine:
+    name = "local"
+
+    def __init__(self, flat, codec, comm, grad_scale=1.0):
+        self.flat = flat
+        self.codec = codec
+        self.gscale = grad_scale
+
+    def step(self, apply_fn, metrics):
+        flat = self.flat
+        for b in flat.buckets:
+            with metrics.timer("decode_time"):
+                ops.reduce_accum(flat.agg_view(b), [flat.grad_view(b)],
+                                 scale=self.gscale, beta=0.0)
+            with metrics.timer("optim_step_time"):
+                apply_fn(b)
+        metrics.add("msg_bytes", 0)
+
+    def finish(self):
+        pass
+
+
+class ReplicatedEngine:
+    name = "replicated"
+
+    def __init__(self, flat, codec, comm, grad_scale=1.0):
+        self.flat = flat
+        self.codec = codec
+        self.comm = comm
+        self.gscale = grad_scale
+        self.use_allreduce = getattr(codec, "supports_allreduce", False)
+        if not self.use_allreduce:
+            dev = flat.flat_param.device
+            self.wire_send = {}
+            self.wire_slots = {}
+            for b in flat.buckets:
+                wn = codec.wire_numel(b.numel, flat.dtype) \
+                    if codec.name == "topk" else codec.wire_numel(b.numel)
+                wd = codec.wire_dtype(flat.dtype)
+                self.wire_send[b.idx] = torch.zeros(wn, dtype=wd, device=dev)
+                self.wire_slots[b.idx] = [
+                    torch.zeros(wn, dtype=wd, device=dev)
+                    for _ in range(comm.world)
+                ]
+
+    def step(self, apply_fn, metrics):
+        flat, codec, comm = self.flat, self.codec, self.comm
+        works = []
+        if self.use_allreduce:
+            with metrics.timer("isend_time"):
+                for b in flat.buckets:
+                    works.append(dist.all_reduce(flat.grad_view(b),
+                                                 async_op=True))
+            nbytes = flat.total * flat.dtype.itemsize
+            for b, w in zip(flat.buckets, works):
+                with metrics.timer("comm_wait"):
+                    w.wait()
+                with metrics.timer("decode_time"):
+                    # all-reduce already summed across ranks -> cast+scale
+                    ops.reduce_accum(flat.agg_view(b), [flat.grad_view(b)],
+                                     scale=self.gscale, beta=0.0)
+                with metrics.timer("optim_step_time"):
+                    apply_fn(b)
+        else:
+            with metrics.timer("code_wait"):
+                for b in flat.buckets:
+                    codec.encode(flat.grad_view(b), self.wire_send[b.idx])
+            with metrics.timer("isend_time"):
+                for b in flat.buckets:
+                    works.append(dist.all_gather(self.wire_slots[b.idx],
+                                                 self.wire_send[b.idx],
+                                                 async_op=True))
+            nbytes = sum(w.numel() * w.dtype.itemsize
+                         for w in self.wire_send.values())
+            for b, w in zip(flat.buckets, works):
+                with metrics.timer("comm_wait"):
+                    w.wait()
+                with metrics.timer("decode_time"):
+                    codec.decode_reduce(flat.agg_view(b),
+                                        self.wire_slots[b.idx],
+                                        gscale=self.gscale, beta=0.0)
+                with metrics.timer("optim_step_time"):
+                    apply_fn(b)
+        metrics.add("msg_bytes", nbytes)
+
+    def finish(self):
+        pass
+
+
+class SyncPSEngine:
+    name = "sync_ps"
+
+    def __init__(self, flat, codec, comm, grad_scale=1.0):
+        self.flat = flat
+        self.codec = codec
+        self.comm = comm
+        self.gscale = grad_scale
+        dev = flat.flat_param.device
+        ident = codec.name == "identity"
+        self.ident = ident
+        self.wire_send = {}
+        self.wire_slots = {}
+        for b in flat.buckets:
+            if ident:
+                self.wire_send[b.idx] = None  # grad view used directly
+            else:
+                wn = codec.wire_numel(b.numel, flat.dtype) \
+                    if codec.name == "topk" else codec.wire_numel(b.numel)
+                wd = codec.wire_dtype(flat.dtype)
+                self.wire_send[b.idx] = torch.zeros(wn, dtype=wd, device=dev)
+            if comm.is_ps:
+                ref = self.wire_send[b.idx]
+                if ref is None:
+                    self.wire_slots[b.idx] = [
+                        torch.zeros(b.numel, dtype=flat.dtype, device=dev)
+                        for _ in range(comm.world)
+                    ]
+                else:
+                    self.wire_slots[b.idx] = [torch.zeros_like(ref)
+                                              for _ in range(comm.world)]
+
+    def _send_wire(self, b):
+        if self.ident:
+            return self.flat.grad_view(b)
+        return self.wire_send[b.idx]
+
+    def step(self, apply_fn, metrics):
+        flat, codec, comm = self.flat, self.codec, self.comm
+        if not self.ident:
+            with metrics.timer("code_wait"):
+                for b in flat.buckets:
+                    codec.encode(flat.grad_view(b), self.wire_send[b.idx])
+        works = []
+        with metrics.timer("isend_time"):
+            for b in flat.buckets:
+                gl = self.wire_slots[b.idx] if comm.is_ps else None
+                works.append(dist.gather(self._send_wire(b), gather_list=gl,
+                                         dst=comm.ps_rank, async_op=True))
+        bworks = []
+        for b, w in zip(flat.buckets, works):
+            with metrics.timer("comm_wait"):
+                w.wait()
+            if comm.is_ps:
+                with metrics.timer("decode_time"):
+                    if self.ident:
+                        codec.decode_reduce(flat.agg_view(b),
+                                            self.wire_slots[b.idx],
+                                            gscale=self.gscale, beta=0.0)
+                    else:
+                        codec.decode_reduce(flat.agg_view(b),
+                                            self.wire_slots[b.idx],
+                                            gscale=self.gscale, beta=0.0)
+                with metrics.timer("optim_step_time"):
+                    apply_fn(b)
+            bworks.append(dist.broadcast(flat.param_view(b), src=comm.ps_rank,
+                                         async_op=True))
+        with metrics.timer("comm_wait"):
+            for w in bworks:
+                w.wait()
+        wire_b = flat.total * flat.dtype.itemsize if self.ident else \
+            sum(t.numel() * t.dtype.itemsize
+                for t in self.wire_send.values())
+        metrics.add("msg_bytes", wire_b + flat.total * flat.dtype.itemsize)
+
+    def finish(self):
+        pass

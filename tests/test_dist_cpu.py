@@ -1,0 +1,202 @@
+"""Multi-process CPU tests (gloo, world_size=2) — BASELINE config 1.
+
+Asserts the SURVEY §4 gap-closing properties:
+  (a) codec round-trip / cross-rank grad-sum equals single-process
+      full-batch gradient (grad_scale='mean', equal shards),
+  (b) replicated params stay bitwise identical across ranks after K steps,
+  (c) sync-PS broadcast keeps ranks identical,
+  (d) async AsySG-InCon trains with bounded staleness and clean shutdown.
+"""
+
+import os
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+pytestmark = pytest.mark.timeout(300)
+
+WORLD = 2
+
+
+def _free_port():
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _setup(rank, world, port):
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["LOCAL_RANK"] = str(rank)
+    from pytorch_ps_mpi_amd import init_distributed
+    return init_distributed(backend="gloo")
+
+
+def _mlp_and_data(rank, batch=16):
+    from pytorch_ps_mpi_amd import models
+    torch.manual_seed(0)
+    model = models.build_model("mlp")
+    # full deterministic batch; each rank takes its shard
+    x, y = models.synthetic_batch("mlp", batch * WORLD, seed=42)
+    xs = x[rank * batch:(rank + 1) * batch]
+    ys = y[rank * batch:(rank + 1) * batch]
+    return model, (x, y), (xs, ys)
+
+
+def _single_process_reference(steps, lr, momentum, batch=16):
+    """torch.optim.SGD on the FULL batch — the grad-sum oracle."""
+    from pytorch_ps_mpi_amd import models
+    torch.manual_seed(0)
+    model = models.build_model("mlp")
+    x, y = models.synthetic_batch("mlp", batch * WORLD, seed=42)
+    opt = torch.optim.SGD(model.parameters(), lr=lr, momentum=momentum)
+    for _ in range(steps):
+        opt.zero_grad()
+        models.loss_fn("mlp", model, x, y).backward()
+        opt.step()
+    return torch.cat([p.detach().flatten() for p in model.parameters()])
+
+
+def _checksums_equal(opt):
+    import torch.distributed as dist
+    cs = [None] * WORLD
+    dist.all_gather_object(cs, opt.flat.param_checksum())
+    return len(set(cs)) == 1
+
+
+def _replicated_worker(rank, port, codec, out_file):
+    from pytorch_ps_mpi_amd import SGD, models
+    _setup(rank, WORLD, port)
+    model, _full, (xs, ys) = _mlp_and_data(rank)
+    opt = SGD(model.named_parameters(), lr=0.05, momentum=0.9,
+              mode="replicated", code=codec, grad_scale="mean",
+              bucket_mb=0.05)
+    for step in range(5):
+        opt.zero_grad()
+        loss = models.loss_fn("mlp", model, xs, ys)
+        loss.backward()
+        opt.step(loss=loss)
+        assert _checksums_equal(opt), f"rank divergence at step {step}"
+    if rank == 0:
+        if codec is None:
+            ref = _single_process_reference(5, lr=0.05, momentum=0.9)
+            got = torch.cat([p.detach().flatten() for p in model.parameters()])
+            err = (got - ref).abs().max().item()
+            assert err < 1e-5, f"grad-sum mismatch vs single-process: {err}"
+        with open(out_file, "w") as f:
+            f.write("ok")
+    opt.finish()
+
+
+def _sync_ps_worker(rank, port, codec, out_file):
+    from pytorch_ps_mpi_amd import SGD, models
+    _setup(rank, WORLD, port)
+    model, _full, (xs, ys) = _mlp_and_data(rank)
+    opt = SGD(model.named_parameters(), lr=0.05, momentum=0.9, mode="ps",
+              code=codec, grad_scale="mean", bucket_mb=0.05)
+    losses = []
+    for step in range(6):
+        opt.zero_grad()
+        loss = models.loss_fn("mlp", model, xs, ys)
+        loss.backward()
+        l, m = opt.step(loss=loss)
+        losses.append(float(l.detach()))
+        assert _checksums_equal(opt), f"rank divergence at step {step}"
+    assert losses[-1] < losses[0]
+    if rank == 0:
+        with open(out_file, "w") as f:
+            f.write("ok")
+    opt.finish()
+
+
+def _async_worker(rank, port, codec, out_file):
+    from pytorch_ps_mpi_amd import SGD, models
+    _setup(rank, WORLD, port)
+    model, _full, (xs, ys) = _mlp_and_data(rank)
+    opt = SGD(model.named_parameters(), lr=0.02, momentum=0.9, mode="async",
+              code=codec, bucket_mb=0.05, window=2, max_stale=4)
+    losses = []
+    for step in range(10):
+        opt.zero_grad()
+        loss = models.loss_fn("mlp", model, xs, ys)
+        loss.backward()
+        l, m = opt.step(loss=loss)
+        losses.append(float(l.detach()))
+        if rank != 0:
+            assert m.get("staleness", 0) <= 4 + 2
+    opt.finish()
+    assert all(torch.isfinite(torch.tensor(losses)))
+    assert losses[-1] < losses[0] * 1.5  # training, not diverging
+    if rank == 0:
+        hist = opt.engine.staleness_hist
+        assert sum(hist.values()) >= 5  # PS actually served pushes
+        with open(out_file, "w") as f:
+            f.write("ok")
+
+
+def _async_dedicated_worker(rank, port, codec, out_file):
+    from pytorch_ps_mpi_amd import SGD, models
+    _setup(rank, WORLD, port)
+    model, _full, (xs, ys) = _mlp_and_data(rank)
+    opt = SGD(model.named_parameters(), lr=0.02, momentum=0.9, mode="async",
+              code=codec, bucket_mb=0.05, window=2, max_stale=4,
+              dedicated_ps=True)
+    if rank == 0:
+        opt.serve()
+        opt.finish()
+        hist = opt.engine.staleness_hist
+        assert sum(hist.values()) >= 8
+        with open(out_file, "w") as f:
+            f.write("ok")
+    else:
+        for step in range(8):
+            opt.zero_grad()
+            loss = models.loss_fn("mlp", model, xs, ys)
+            loss.backward()
+            opt.step(loss=loss)
+        opt.finish()
+
+
+def _spawn(fn, codec, tmp_path):
+    out = str(tmp_path / "ok.txt")
+    port = _free_port()
+    mp.spawn(fn, args=(port, codec, out), nprocs=WORLD, join=True)
+    assert os.path.exists(out)
+
+
+def test_replicated_identity(tmp_path):
+    _spawn(_replicated_worker, None, tmp_path)
+
+
+def test_replicated_quant8(tmp_path):
+    _spawn(_replicated_worker, "quant8", tmp_path)
+
+
+def test_replicated_topk(tmp_path):
+    _spawn(_replicated_worker, "topk:0.25", tmp_path)
+
+
+def test_sync_ps_identity(tmp_path):
+    _spawn(_sync_ps_worker, None, tmp_path)
+
+
+def test_sync_ps_quant8(tmp_path):
+    _spawn(_sync_ps_worker, "quant8", tmp_path)
+
+
+def test_async_colocated(tmp_path):
+    _spawn(_async_worker, None, tmp_path)
+
+
+def test_async_topk(tmp_path):
+    _spawn(_async_worker, "topk:0.25", tmp_path)
+
+
+def test_async_dedicated_ps(tmp_path):
+    _spawn(_async_dedicated_worker, None, tmp_path)

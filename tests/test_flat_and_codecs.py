@@ -1,0 +1,98 @@
+import torch
+
+from pytorch_ps_mpi_amd import codecs, models
+from pytorch_ps_mpi_amd.utils.flat import ALIGN, FlatSpace
+
+
+def test_flatspace_views_and_buckets():
+    m = models.build_model("mlp")
+    flat = FlatSpace(m.named_parameters(), bucket_elems=1000)
+    # params are views into flat_param
+    for name, p, o, n in flat.entries:
+        assert p.data.data_ptr() == flat.flat_param[o:o + n].data_ptr()
+        assert o % ALIGN == 0
+    assert flat.total % ALIGN == 0
+    assert len(flat.buckets) >= 2
+    assert flat.buckets[0].start == 0
+    assert flat.buckets[-1].end == flat.total
+    for a, b in zip(flat.buckets, flat.buckets[1:]):
+        assert a.end == b.start
+    # grads accumulate into flat_grad
+    x, y = models.synthetic_batch("mlp", 4, seed=0)
+    loss = models.loss_fn("mlp", m, x, y)
+    loss.backward()
+    assert flat.flat_grad.abs().sum() > 0
+    for name, p, o, n in flat.entries:
+        assert p.grad.data_ptr() == flat.flat_grad[o:o + n].data_ptr()
+    flat.zero_grad()
+    assert flat.flat_grad.abs().sum() == 0
+
+
+def test_flatspace_reattach_after_set_to_none():
+    m = models.build_model("mlp")
+    flat = FlatSpace(m.named_parameters(), bucket_elems=1 << 20)
+    for p in m.parameters():
+        p.grad = None
+    flat.zero_grad()
+    x, y = models.synthetic_batch("mlp", 4, seed=0)
+    models.loss_fn("mlp", m, x, y).backward()
+    assert flat.flat_grad.abs().sum() > 0
+
+
+def _codec_roundtrip(codec, n=4096, tol=None):
+    torch.manual_seed(0)
+    src = torch.randn(n)
+    wn = codec.wire_numel(n, torch.float32) if codec.name == "topk" \
+        else codec.wire_numel(n)
+    wire = torch.zeros(wn, dtype=codec.wire_dtype(torch.float32))
+    codec.encode(src, wire)
+    dst = torch.zeros(n)
+    codec.decode_reduce(dst, [wire], src_dtype=torch.float32)
+    return src, dst
+
+
+def test_identity_codec():
+    c = codecs.Identity()
+    src, dst = _codec_roundtrip(c)
+    assert torch.allclose(src, dst)
+
+
+def test_quant8_codec():
+    c = codecs.QuantInt8()
+    src, dst = _codec_roundtrip(c)
+    assert (dst - src).abs().max() < 0.05  # half-step of absmax/127 chunks
+
+
+def test_topk_codec():
+    c = codecs.TopK(density=0.1)
+    src, dst = _codec_roundtrip(c)
+    k = c.k_for(src.numel())
+    nz = (dst != 0).sum().item()
+    assert nz == k
+    mask = dst != 0
+    assert torch.allclose(dst[mask], src[mask])
+
+
+def test_topk_multi_message_sum():
+    c = codecs.TopK(density=0.5)
+    n = 1024
+    torch.manual_seed(1)
+    a, b = torch.randn(n), torch.randn(n)
+    wn = c.wire_numel(n, torch.float32)
+    wa = torch.zeros(wn, dtype=torch.uint8)
+    wb = torch.zeros(wn, dtype=torch.uint8)
+    c.encode(a, wa)
+    c.encode(b, wb)
+    dst = torch.zeros(n)
+    c.decode_reduce(dst, [wa, wb], gscale=0.5, src_dtype=torch.float32)
+    # where both picked the same index the result is the scaled sum
+    k = c.k_for(n)
+    assert (dst != 0).sum() >= k  # union of supports
+
+
+def test_get_codec_spec():
+    assert codecs.get_codec(None).name == "identity"
+    assert codecs.get_codec("topk:0.05").density == 0.05
+    assert codecs.get_codec("quant8").name == "quant8"
+    c = codecs.TopK()
+    assert codecs.get_codec(c) is c

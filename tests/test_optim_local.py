@@ -1,0 +1,136 @@
+"""Single-process optimizer semantics: parity with torch.optim, checkpointing,
+lr-scheduler compatibility, reference API shape (step -> (loss, metrics))."""
+
+import torch
+
+from pytorch_ps_mpi_amd import SGD, Adam, models
+from pytorch_ps_mpi_amd.utils import checkpoint
+
+
+def _train(model, opt, steps=5, seed=1):
+    x, y = models.synthetic_batch("mlp", 32, seed=seed)
+    losses = []
+    for _ in range(steps):
+        opt.zero_grad()
+        loss = models.loss_fn("mlp", model, x, y)
+        loss.backward()
+        l, metrics = opt.step(loss=loss)
+        losses.append(float(l.detach()))
+    return losses, metrics
+
+
+def _torch_train(model, opt, steps=5, seed=1):
+    x, y = models.synthetic_batch("mlp", 32, seed=seed)
+    for _ in range(steps):
+        opt.zero_grad()
+        loss = models.loss_fn("mlp", model, x, y)
+        loss.backward()
+        opt.step()
+
+
+def _params(m):
+    return torch.cat([p.detach().flatten() for p in m.parameters()])
+
+
+def test_sgd_parity_with_torch():
+    torch.manual_seed(0)
+    m1 = models.build_model("mlp")
+    torch.manual_seed(0)
+    m2 = models.build_model("mlp")
+    opt1 = SGD(m1.named_parameters(), lr=0.1, momentum=0.9, weight_decay=1e-4)
+    opt2 = torch.optim.SGD(m2.parameters(), lr=0.1, momentum=0.9,
+                           weight_decay=1e-4)
+    _train(m1, opt1)
+    _torch_train(m2, opt2)
+    assert torch.allclose(_params(m1), _params(m2), atol=1e-6)
+
+
+def test_adam_parity_with_torch():
+    torch.manual_seed(0)
+    m1 = models.build_model("mlp")
+    torch.manual_seed(0)
+    m2 = models.build_model("mlp")
+    opt1 = Adam(m1.named_parameters(), lr=1e-3, betas=(0.9, 0.999),
+                weight_decay=1e-2, amsgrad=True)
+    opt2 = torch.optim.Adam(m2.parameters(), lr=1e-3, betas=(0.9, 0.999),
+                            weight_decay=1e-2, amsgrad=True)
+    _train(m1, opt1)
+    _torch_train(m2, opt2)
+    assert torch.allclose(_params(m1), _params(m2), atol=1e-5)
+
+
+def test_step_returns_loss_and_metrics():
+    torch.manual_seed(0)
+    m = models.build_model("mlp")
+    opt = SGD(m.named_parameters(), lr=0.1)
+    losses, metrics = _train(m, opt, steps=3)
+    assert losses[-1] < losses[0]
+    assert "optim_step_time" in metrics and "step" in metrics
+
+
+def test_closure_api():
+    torch.manual_seed(0)
+    m = models.build_model("mlp")
+    opt = SGD(m.named_parameters(), lr=0.1)
+    x, y = models.synthetic_batch("mlp", 8, seed=2)
+
+    def closure():
+        opt.zero_grad()
+        loss = models.loss_fn("mlp", m, x, y)
+        loss.backward()
+        return loss
+
+    loss, metrics = opt.step(closure)
+    assert loss is not None
+
+
+def test_lr_scheduler_compat():
+    torch.manual_seed(0)
+    m = models.build_model("mlp")
+    opt = SGD(m.named_parameters(), lr=0.1, momentum=0.9)
+    sched = torch.optim.lr_scheduler.StepLR(opt, step_size=1, gamma=0.5)
+    _train(m, opt, steps=1)
+    sched.step()
+    assert abs(opt.param_groups[0]["lr"] - 0.05) < 1e-9
+
+
+def test_checkpoint_resume(tmp_path):
+    torch.manual_seed(0)
+    m1 = models.build_model("mlp")
+    opt1 = SGD(m1.named_parameters(), lr=0.1, momentum=0.9)
+    _train(m1, opt1, steps=3)
+    path = tmp_path / "ck.pt"
+    checkpoint.save(str(path), opt1, extra={"epoch": 3})
+
+    torch.manual_seed(123)  # different init; must be overwritten by load
+    m2 = models.build_model("mlp")
+    opt2 = SGD(m2.named_parameters(), lr=0.1, momentum=0.9)
+    extra = checkpoint.load(str(path), opt2)
+    assert extra["epoch"] == 3
+    assert torch.allclose(_params(m1), _params(m2))
+    # continued training must match exactly
+    _train(m1, opt1, steps=2, seed=9)
+    _train(m2, opt2, steps=2, seed=9)
+    assert torch.allclose(_params(m1), _params(m2))
+
+
+def test_models_forward_backward_small():
+    for name, kw, batch in [
+        ("resnet18", {"num_classes": 10}, 2),
+        ("vit_b16", {"num_classes": 10}, 1),
+    ]:
+        m = models.build_model(name, **kw)
+        x, y = models.synthetic_batch(name, batch, seed=0)
+        y = y % 10
+        loss = torch.nn.functional.cross_entropy(m(x), y)
+        loss.backward()
+        assert torch.isfinite(loss)
+
+
+def test_gpt2_tiny_forward_backward():
+    from pytorch_ps_mpi_amd.models.gpt2 import GPT2
+    m = GPT2(vocab=128, ctx=32, dim=64, depth=2, heads=2)
+    x = torch.randint(0, 128, (2, 17))
+    loss = m.loss(x[:, :-1], x[:, 1:])
+    loss.backward()
+    assert torch.isfinite(loss)
