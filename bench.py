@@ -1,0 +1,167 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: ResNet-50 async-PS training throughput on MI355X.
+
+BASELINE.json metric: "SGD steps/sec + samples/sec, ResNet-50 async-PS at
+1/2/4/8 MI355X".  Synthetic ImageNet-shaped data, random-init weights, bf16
+compute with fp32 master (the reference publishes no numbers — BASELINE.md —
+so this harness IS the baseline series).
+
+Run (driver contract):
+  python bench.py --gpus N --steps K --warmup W
+  torchrun --nnodes=1 --nproc-per-node N --master-addr 127.0.0.1 bench.py ...
+
+Per-GPU work is fixed (weak scaling).  Rank 0 prints ONE JSON line.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--model", default="resnet50")
+    ap.add_argument("--batch", type=int, default=64,
+                    help="per-rank batch size")
+    ap.add_argument("--mode", default="async",
+                    choices=["async", "replicated", "ps"])
+    ap.add_argument("--codec", default=None,
+                    help="identity|topk:D|quant8 (default identity)")
+    ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    ap.add_argument("--optim", default="sgd", choices=["sgd", "adam"])
+    ap.add_argument("--lr", type=float, default=0.0125)
+    ap.add_argument("--bucket-mb", type=float, default=50.0)
+    ap.add_argument("--dedicated-ps", action="store_true")
+    ap.add_argument("--window", type=int, default=2)
+    ap.add_argument("--max-stale", type=int, default=8)
+    ap.add_argument("--seq-len", type=int, default=512)
+    return ap.parse_args()
+
+
+def main():
+    args = parse_args()
+    import torch.distributed as dist
+
+    from pytorch_ps_mpi_amd import SGD, Adam, init_distributed, models
+
+    device = init_distributed()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    on_gpu = device.type == "cuda"
+    dtype = torch.bfloat16 if (args.dtype == "bf16" and on_gpu) else \
+        torch.float32
+
+    if on_gpu:
+        from pytorch_ps_mpi_amd import ops
+        assert ops.HAVE_EXT, "HIP extension must be loaded on GPU"
+
+    torch.manual_seed(1234)
+    model = models.build_model(args.model, device=device, dtype=dtype)
+    opt_cls = {"sgd": SGD, "adam": Adam}[args.optim]
+    opt_kw = dict(mode=args.mode, code=args.codec,
+                  bucket_mb=args.bucket_mb, grad_scale="mean",
+                  window=args.window, max_stale=args.max_stale,
+                  dedicated_ps=args.dedicated_ps)
+    if args.optim == "sgd":
+        opt = opt_cls(model.named_parameters(), lr=args.lr, momentum=0.9,
+                      **opt_kw)
+    else:
+        opt = opt_cls(model.named_parameters(), lr=args.lr, **opt_kw)
+
+    x, y = models.synthetic_batch(args.model, args.batch, device=device,
+                                  dtype=dtype, seed=100 + rank,
+                                  seq_len=args.seq_len)
+
+    is_serving_ps = (args.mode == "async" and args.dedicated_ps and
+                     world > 1 and rank == 0)
+    n_train = world - 1 if (args.mode == "async" and args.dedicated_ps
+                            and world > 1) else world
+
+    def one_step():
+        opt.zero_grad()
+        loss = models.loss_fn(args.model, model, x, y)
+        loss.backward()
+        opt.step(loss=loss)
+        return loss
+
+    def barrier_sync():
+        if dist.is_initialized():
+            dist.barrier()
+        if on_gpu:
+            torch.cuda.synchronize()
+
+    if is_serving_ps:
+        # dedicated PS: no barriers (workers would stall waiting for
+        # replies) — serve until every worker sent its stop marker.
+        opt.serve()
+        opt.finish()
+        elapsed = torch.tensor([0.0], dtype=torch.float64)
+    else:
+        for _ in range(args.warmup):
+            one_step()
+        if not args.dedicated_ps:
+            barrier_sync()
+        elif on_gpu:
+            torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(args.steps):
+            one_step()
+        if on_gpu:
+            torch.cuda.synchronize()
+        t1 = time.perf_counter()
+        elapsed = torch.tensor([t1 - t0], dtype=torch.float64)
+        if args.mode == "async":
+            opt.finish()  # drain replies, send stop (before any collective)
+
+    # max over training ranks (dedicated PS contributes 0)
+    if dist.is_initialized():
+        dist.all_reduce(elapsed, op=dist.ReduceOp.MAX)
+    barrier_sync()
+
+    t = float(elapsed.item())
+    steps_per_sec = args.steps / t
+    samples_per_sec = n_train * args.batch * args.steps / t
+
+    if rank == 0:
+        out = {
+            "metric": "samples_per_sec",
+            "value": samples_per_sec,
+            "unit": "samples/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": 1000.0 * t / args.steps,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if dtype == torch.bfloat16 else "fp32",
+            "data": "synthetic",
+            "steps_per_sec": steps_per_sec,
+            "config": {
+                "model": args.model,
+                "global_batch": n_train * args.batch,
+                "parallelism": (
+                    f"async-ps({'dedicated' if args.dedicated_ps else 'colocated'},"
+                    f"dp{world})" if args.mode == "async"
+                    else f"{args.mode}(dp{world})"),
+                "optim": args.optim,
+                "codec": opt.codec.name,
+                "bucket_mb": args.bucket_mb,
+            },
+        }
+        print(json.dumps(out))
+
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
