@@ -43,6 +43,8 @@ def parse_args():
     ap.add_argument("--window", type=int, default=2)
     ap.add_argument("--max-stale", type=int, default=8)
     ap.add_argument("--seq-len", type=int, default=512)
+    ap.add_argument("--no-channels-last", action="store_true",
+                    help="disable NHWC layout for conv models")
     return ap.parse_args()
 
 
@@ -64,7 +66,14 @@ def main():
         assert ops.HAVE_EXT, "HIP extension must be loaded on GPU"
 
     torch.manual_seed(1234)
+    if on_gpu:
+        torch.backends.cudnn.benchmark = True  # MIOpen find mode
     model = models.build_model(args.model, device=device, dtype=dtype)
+    image_model = args.model in ("resnet18", "resnet50", "vit_b16")
+    channels_last = (on_gpu and image_model and not args.no_channels_last
+                     and args.model != "vit_b16")
+    if channels_last:
+        model = model.to(memory_format=torch.channels_last)
     opt_cls = {"sgd": SGD, "adam": Adam}[args.optim]
     opt_kw = dict(mode=args.mode, code=args.codec,
                   bucket_mb=args.bucket_mb, grad_scale="mean",
@@ -79,6 +88,8 @@ def main():
     x, y = models.synthetic_batch(args.model, args.batch, device=device,
                                   dtype=dtype, seed=100 + rank,
                                   seq_len=args.seq_len)
+    if channels_last:
+        x = x.contiguous(memory_format=torch.channels_last)
 
     is_serving_ps = (args.mode == "async" and args.dedicated_ps and
                      world > 1 and rank == 0)

@@ -82,11 +82,24 @@ class FlatSpace:
         else:
             self.master = torch.zeros(self.total, dtype=torch.float32, device=dev)
 
+        # channels_last conv weights keep their NHWC layout: the flat slice
+        # stores NHWC-contiguous bytes and the param view is a permute of it.
+        self._chlast = set()
+        for name, p, o, n in self.entries:
+            if (p.dim() == 4 and
+                    p.data.is_contiguous(memory_format=torch.channels_last)
+                    and not p.data.is_contiguous()):
+                self._chlast.add(name)
+
         with torch.no_grad():
             for name, p, o, n in self.entries:
-                self.flat_param[o:o + n].copy_(
-                    p.data.reshape(-1).to(device=dev, dtype=self.dtype))
-                p.data = self.flat_param[o:o + n].view(p.shape)
+                src = p.data.to(device=dev, dtype=self.dtype)
+                if name in self._chlast:
+                    src = src.permute(0, 2, 3, 1).reshape(-1)
+                else:
+                    src = src.reshape(-1)
+                self.flat_param[o:o + n].copy_(src)
+                p.data = self._shaped_view(self.flat_param, name, p.shape, o, n)
             if self.master is not self.flat_param:
                 self.master.copy_(self.flat_param.float())
         self.attach_grads()
@@ -126,16 +139,23 @@ class FlatSpace:
 
     # ---- grad management -------------------------------------------------
 
+    def _shaped_view(self, buf, name, shape, o, n):
+        if name in self._chlast:
+            N, C, H, W = shape
+            return buf[o:o + n].view(N, H, W, C).permute(0, 3, 1, 2)
+        return buf[o:o + n].view(shape)
+
     def attach_grads(self):
         for name, p, o, n in self.entries:
-            p.grad = self.flat_grad[o:o + n].view(p.shape)
+            p.grad = self._shaped_view(self.flat_grad, name, p.shape, o, n)
 
     def zero_grad(self):
         self.flat_grad.zero_()
         # re-attach in case user code dropped the views (set_to_none etc.)
         for name, p, o, n in self.entries:
-            if p.grad is None or p.grad.data_ptr() != self.flat_grad[o:o + n].data_ptr():
-                p.grad = self.flat_grad[o:o + n].view(p.shape)
+            if p.grad is None or \
+                    p.grad.data_ptr() != self.flat_grad[o:o + n].data_ptr():
+                p.grad = self._shaped_view(self.flat_grad, name, p.shape, o, n)
 
     def sync_param_from_master(self):
         """flat_param (model dtype) <- master (fp32)."""

@@ -96,3 +96,32 @@ def test_get_codec_spec():
     assert codecs.get_codec("quant8").name == "quant8"
     c = codecs.TopK()
     assert codecs.get_codec(c) is c
+
+
+def test_flatspace_channels_last():
+    import torch.nn as nn
+    torch.manual_seed(0)
+    m = nn.Sequential(nn.Conv2d(3, 8, 3, padding=1), nn.ReLU(),
+                      nn.Conv2d(8, 4, 1))
+    m = m.to(memory_format=torch.channels_last)
+    ref = {k: v.detach().clone() for k, v in m.named_parameters()}
+    flat = FlatSpace(m.named_parameters(), bucket_elems=1 << 20)
+    for name, p in m.named_parameters():
+        assert torch.equal(p.detach(), ref[name]), name
+        if p.dim() == 4:
+            assert p.is_contiguous(memory_format=torch.channels_last)
+    x = torch.randn(2, 3, 8, 8).contiguous(memory_format=torch.channels_last)
+    y = m(x).sum()
+    y.backward()
+    assert flat.flat_grad.abs().sum() > 0
+    # grads land in the flat buffer through the permuted views
+    for name, p, o, n in flat.entries:
+        assert p.grad.data_ptr() == flat.flat_grad[o:o + n].data_ptr()
+    # grad values match a plain-layout reference model
+    torch.manual_seed(0)
+    m2 = nn.Sequential(nn.Conv2d(3, 8, 3, padding=1), nn.ReLU(),
+                       nn.Conv2d(8, 4, 1))
+    m2(x.contiguous()).sum().backward()
+    for (n1, p1), (n2, p2) in zip(m.named_parameters(),
+                                  m2.named_parameters()):
+        assert torch.allclose(p1.grad, p2.grad, atol=1e-5), n1
