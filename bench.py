@@ -26,10 +26,10 @@ import torch
 def parse_args():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=20)
-    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--model", default="resnet50")
-    ap.add_argument("--batch", type=int, default=64,
+    ap.add_argument("--batch", type=int, default=1024,
                     help="per-rank batch size")
     ap.add_argument("--mode", default="async",
                     choices=["async", "replicated", "ps"])
@@ -52,6 +52,10 @@ def parse_args():
 
 def main():
     args = parse_args()
+    if os.environ.get("BENCH_DEBUG_HANG"):
+        import faulthandler
+        faulthandler.dump_traceback_later(
+            int(os.environ["BENCH_DEBUG_HANG"]), exit=True)
     import torch.distributed as dist
 
     from pytorch_ps_mpi_amd import SGD, Adam, init_distributed, models
@@ -123,7 +127,10 @@ def main():
     else:
         for _ in range(args.warmup):
             one_step()
-        if not args.dedicated_ps:
+        # async mode: NO barrier here — the colocated PS must keep serving
+        # while workers run (a PS parked in a collective deadlocks a worker
+        # that hit its staleness/window bound).  Warmup aligns the ranks.
+        if args.mode != "async":
             barrier_sync()
         elif on_gpu:
             torch.cuda.synchronize()

@@ -3,21 +3,25 @@
 Semantics (arXiv:1506.08272, "inconsistent reads"): workers compute gradients
 on whatever parameter version they last received and push them to the PS
 without stalling; the PS applies each push as it arrives (or per `quorum`
-pushes) and replies with a current parameter snapshot.  Staleness is bounded
-by the worker-side window: a worker blocks only when `max_stale` of its
-pushes are still unanswered.
+pushes) and replies with a current parameter snapshot.
 
-MI355X mapping: one process per GPU; each (PS, worker) pair has its OWN
-process group, so its RCCL sends/recvs ride a dedicated xGMI p2p channel and
-order independently of other peers (the PS drains all 7 peers concurrently —
-the reference instead polled MPI ANY_SOURCE on the host).  All payloads are
-flat device tensors; no host round trip.
+MI355X mapping — the device-side gradient ring buffer of SURVEY §2.3:
+  * one process per GPU; each (PS, worker) pair gets TWO process groups
+    (push channel / reply channel), so each channel's RCCL p2p ops order
+    independently and ride a dedicated xGMI link pair;
+  * the PS keeps a ring of `ring` pre-posted receive slots PER PEER in HBM —
+    workers run ahead of the PS's serve cadence up to the ring depth and
+    never stall on the PS (the reference instead polled MPI ANY_SOURCE on
+    the host with pickled payloads);
+  * all payloads are flat device tensors; no host round trips.
 
-Message protocol per pair (fixed sizes, fixed per-pair order):
-  worker -> PS : hdr int64[2] = (worker_step, param_version_used), wire
-  PS -> worker : hdr int64[1] = (ps_version,), param snapshot (model dtype)
-A worker sends hdr=(-1,-1) + dummy wire to stop; the PS stops serving a peer
-after its stop and finish() returns when all peers stopped.
+Message protocol (fixed sizes; per-channel order is the matching order):
+  push  : hdr int64[2] = (worker_step, param_version_used)  +  wire
+  reply : hdr int64[1] = (ps_version,)  +  flat_param snapshot (model dtype)
+Workers post exactly one reply-recv pair per push (1:1), so every posted
+recv is eventually matched.  To stop, a worker sends `ring` stop markers
+(hdr=(-1,-1) + dummy wire) — one for every recv slot the PS keeps posted —
+and the PS retires that peer without replying.
 
 Colocated mode (default): rank 0 trains too and serves peers opportunistically
 between its own steps.  Dedicated mode: rank 0 only serves (`serve()`).
@@ -31,32 +35,58 @@ import torch.distributed as dist
 from .. import ops
 
 
-class _PeerState:
-    __slots__ = ("rank", "group", "hdr", "wire", "reqs", "reply_hdr",
-                 "reply_buf", "reply_reqs", "stopped")
+def _done(reqs):
+    return all(r.is_completed() for r in reqs)
 
-    def __init__(self, rank, group, hdr, wire, reply_hdr, reply_buf):
+
+def _wait(reqs):
+    for r in reqs:
+        r.wait()
+
+
+class _Peer:
+    """PS-side per-worker state: recv ring + reply ring."""
+
+    __slots__ = ("rank", "push_g", "reply_g", "slots", "head", "replies",
+                 "rhead", "stopped")
+
+    def __init__(self, rank, push_g, reply_g, ring, reply_ring, wn, wdt,
+                 total, pdt, dev):
         self.rank = rank
-        self.group = group
-        self.hdr = hdr
-        self.wire = wire
-        self.reqs = []
-        self.reply_hdr = reply_hdr
-        self.reply_buf = reply_buf
-        self.reply_reqs = []
+        self.push_g = push_g
+        self.reply_g = reply_g
+        self.slots = [{
+            "hdr": torch.zeros(2, dtype=torch.int64, device=dev),
+            "wire": torch.zeros(wn, dtype=wdt, device=dev),
+            "reqs": None,
+        } for _ in range(ring)]
+        self.head = 0
+        self.replies = [{
+            "hdr": torch.zeros(1, dtype=torch.int64, device=dev),
+            "buf": torch.zeros(total, dtype=pdt, device=dev),
+            "reqs": [],
+        } for _ in range(reply_ring)]
+        self.rhead = 0
         self.stopped = False
+
+    def post(self, slot):
+        slot["reqs"] = [
+            dist.irecv(slot["hdr"], src=self.rank, group=self.push_g),
+            dist.irecv(slot["wire"], src=self.rank, group=self.push_g),
+        ]
 
 
 class AsyncPSEngine:
     name = "async_ps"
 
-    def __init__(self, flat, codec, comm, grad_scale=1.0, window=2,
-                 max_stale=8, quorum=1, dedicated=False):
+    def __init__(self, flat, codec, comm, grad_scale=1.0, window=4,
+                 max_stale=8, quorum=1, dedicated=False, reply_ring=2):
         self.flat = flat
         self.codec = codec
         self.comm = comm
         self.gscale = grad_scale
-        self.window = max(1, int(window))
+        self.ring = max(1, int(window))     # PS recv slots per peer
+        self.window = self.ring             # worker in-flight push bound
         self.max_stale = int(max_stale)
         self.quorum = max(1, int(quorum))
         self.dedicated = bool(dedicated)
@@ -82,78 +112,67 @@ class AsyncPSEngine:
             for w in range(comm.world):
                 if w == comm.ps_rank:
                     continue
-                st = _PeerState(
-                    w, comm.pair_group(w),
-                    hdr=torch.zeros(2, dtype=torch.int64, device=dev),
-                    wire=torch.zeros(wn, dtype=self.wire_dtype, device=dev),
-                    reply_hdr=torch.zeros(1, dtype=torch.int64, device=dev),
-                    reply_buf=torch.zeros(total, dtype=flat.dtype, device=dev),
-                )
+                st = _Peer(w, comm.push_group(w), comm.reply_group(w),
+                           self.ring, reply_ring, wn, self.wire_dtype,
+                           total, flat.dtype, dev)
                 self.peers[w] = st
+            # pre-post every recv slot, per peer, in ring order
             for st in self.peers.values():
-                self._post_recv(st)
+                for s in st.slots:
+                    st.post(s)
         else:
-            g = comm.pair_group(comm.rank)
-            self.group = g
-            self.slots = []
-            for _ in range(self.window):
-                self.slots.append({
-                    "hdr": torch.zeros(2, dtype=torch.int64, device=dev),
-                    "wire": torch.zeros(wn, dtype=self.wire_dtype, device=dev),
-                    "phdr": torch.zeros(1, dtype=torch.int64, device=dev),
-                    "pbuf": torch.zeros(total, dtype=flat.dtype, device=dev),
-                    "reqs": None,
-                })
-            self.inflight = []  # slot indices, oldest first
+            self.push_g = comm.push_group(comm.rank)
+            self.reply_g = comm.reply_group(comm.rank)
+            self.pushes = [{
+                "hdr": torch.zeros(2, dtype=torch.int64, device=dev),
+                "wire": torch.zeros(wn, dtype=self.wire_dtype, device=dev),
+                "reqs": None,
+            } for _ in range(self.window)]
+            self.rslots = [{
+                "hdr": torch.zeros(1, dtype=torch.int64, device=dev),
+                "buf": torch.zeros(total, dtype=flat.dtype, device=dev),
+                "reqs": None,
+            } for _ in range(self.window)]
+            self.sent = 0
+            self.harvested = 0
             self.worker_step = 0
             self.param_version = 0
             self.last_applied_step = 0
 
     # ------------------------------------------------------------------ PS
 
-    def _post_recv(self, st):
-        st.reqs = [
-            dist.irecv(st.hdr, src=st.rank, group=st.group),
-            dist.irecv(st.wire, src=st.rank, group=st.group),
-        ]
-
-    def _ps_decode_full(self, wire, beta):
-        self.codec.decode_reduce(self.flat.agg, [wire], gscale=self.gscale,
-                                 beta=beta, src_dtype=self.flat.dtype)
-
-    def poll_serve(self, metrics, block_for=0):
-        """Serve any peers whose push has arrived.  PS-side only."""
-        served = 0
-        for st in self.peers.values():
-            if st.stopped or not st.reqs:
-                continue
-            done = all(r.is_completed() for r in st.reqs)
-            if done or block_for > 0:
-                if self._serve_one(st, metrics):
-                    served += 1
-        return served
-
-    def _serve_one(self, st, metrics):
+    def _serve_slot(self, st, metrics):
+        """Process the head recv slot of peer st (must be completed/waited)."""
         flat, codec = self.flat, self.codec
-        for r in st.reqs:
-            r.wait()
-        st.reqs = []
-        hdr = st.hdr.tolist()
+        slot = st.slots[st.head]
+        _wait(slot["reqs"])
+        slot["reqs"] = None
+        hdr = slot["hdr"].tolist()
         if hdr[0] < 0:
+            # stop marker: drain the remaining posted slots (the worker sends
+            # `ring` markers, one per posted slot), never repost
+            for k in range(1, self.ring):
+                s2 = st.slots[(st.head + k) % self.ring]
+                if s2["reqs"] is not None:
+                    _wait(s2["reqs"])
+                    s2["reqs"] = None
             st.stopped = True
-            return True
+            return
         staleness = max(0, self.ps_version - int(hdr[1]))
         self.staleness_hist[staleness] = \
             self.staleness_hist.get(staleness, 0) + 1
         beta = 1.0 if self._accum_count > 0 else 0.0
-        if codec.name == "identity":
-            for b in flat.buckets:
-                codec.decode_reduce(flat.agg_view(b),
-                                    [st.wire[b.start:b.end]],
+        with metrics.timer("decode_time"):
+            if codec.name == "identity":
+                for b in flat.buckets:
+                    codec.decode_reduce(flat.agg_view(b),
+                                        [slot["wire"][b.start:b.end]],
+                                        gscale=self.gscale, beta=beta,
+                                        src_dtype=flat.dtype)
+            else:
+                codec.decode_reduce(flat.agg, [slot["wire"]],
                                     gscale=self.gscale, beta=beta,
                                     src_dtype=flat.dtype)
-        else:
-            self._ps_decode_full(st.wire, beta)
         self._accum_count += 1
         if self._accum_count >= self.quorum:
             with metrics.timer("optim_step_time"):
@@ -161,93 +180,118 @@ class AsyncPSEngine:
                     self._apply_fn(b)
             self._accum_count = 0
             self.ps_version += 1
-        for r in st.reply_reqs:
-            r.wait()
-        st.reply_hdr.fill_(self.ps_version)
-        st.reply_buf.copy_(flat.flat_param)
-        st.reply_reqs = [
-            dist.isend(st.reply_hdr, dst=st.rank, group=st.group),
-            dist.isend(st.reply_buf, dst=st.rank, group=st.group),
+        # reply with a parameter snapshot on the reply channel
+        rep = st.replies[st.rhead]
+        st.rhead = (st.rhead + 1) % len(st.replies)
+        _wait(rep["reqs"])  # snapshot buffer must be free
+        rep["hdr"].fill_(self.ps_version)
+        rep["buf"].copy_(flat.flat_param)
+        rep["reqs"] = [
+            dist.isend(rep["hdr"], dst=st.rank, group=st.reply_g),
+            dist.isend(rep["buf"], dst=st.rank, group=st.reply_g),
         ]
-        self._post_recv(st)
-        return True
+        # repost this recv slot at the tail of the ring
+        st.post(slot)
+        st.head = (st.head + 1) % self.ring
 
-    def serve(self, metrics, until_all_stopped=True):
-        """Dedicated-PS loop: serve peers until every peer sent a stop."""
+    def poll_serve(self, metrics, max_per_peer=None):
+        """Serve arrived pushes without blocking. PS-side only."""
+        served = 0
+        budget = max_per_peer if max_per_peer is not None else self.ring
+        for st in self.peers.values():
+            n = 0
+            while (not st.stopped and n < budget
+                   and st.slots[st.head]["reqs"] is not None
+                   and _done(st.slots[st.head]["reqs"])):
+                self._serve_slot(st, metrics)
+                served += 1
+                n += 1
+        return served
+
+    def serve(self, metrics):
+        """Dedicated-PS loop: serve until every peer sent its stop."""
         while True:
             alive = [st for st in self.peers.values() if not st.stopped]
             if not alive:
                 break
-            progressed = 0
-            for st in alive:
-                if st.reqs and all(r.is_completed() for r in st.reqs):
-                    self._serve_one(st, metrics)
-                    progressed += 1
+            progressed = self.poll_serve(metrics)
             if progressed == 0:
-                # block on one peer to make progress without spinning hot
-                st = alive[0]
-                self._serve_one(st, metrics)
+                # block on one peer's head slot to make progress
+                self._serve_slot(alive[0], metrics)
 
     # -------------------------------------------------------------- worker
 
-    def _worker_apply_params(self, slot, metrics):
-        """Copy a completed param reply into live params."""
-        with metrics.timer("decode_time"):
-            self.flat.flat_param.copy_(slot["pbuf"])
-            self.param_version = int(slot["phdr"].item())
-            self.last_applied_step = self.worker_step
+    def _apply_reply(self, slot, metrics, skip_copy=False):
+        if not skip_copy:
+            with metrics.timer("decode_time"):
+                self.flat.flat_param.copy_(slot["buf"])
+                self.param_version = int(slot["hdr"].item())
+                self.last_applied_step = self.worker_step
+        slot["reqs"] = None
+        self.harvested += 1
 
-    def worker_step_exchange(self, metrics):
-        """Push the current gradient; harvest any arrived param reply."""
-        flat, codec = self.flat, self.codec
-        self.worker_step += 1
-        slot_idx = self.worker_step % self.window
-        slot = self.slots[slot_idx]
-        if slot["reqs"] is not None:
-            with metrics.timer("comm_wait"):
-                for r in slot["reqs"]:
-                    r.wait()
-            self._worker_apply_params(slot, metrics)
-            slot["reqs"] = None
-            self.inflight = [i for i in self.inflight if i != slot_idx]
-        with metrics.timer("code_wait"):
-            if codec.name == "identity":
-                slot["wire"].copy_(flat.flat_grad)
-            else:
-                codec.encode(flat.flat_grad, slot["wire"])
-        slot["hdr"][0] = self.worker_step
-        slot["hdr"][1] = self.param_version
-        with metrics.timer("isend_time"):
-            slot["reqs"] = [
-                dist.isend(slot["hdr"], dst=self.comm.ps_rank, group=self.group),
-                dist.isend(slot["wire"], dst=self.comm.ps_rank, group=self.group),
-                dist.irecv(slot["phdr"], src=self.comm.ps_rank, group=self.group),
-                dist.irecv(slot["pbuf"], src=self.comm.ps_rank, group=self.group),
-            ]
-        self.inflight.append(slot_idx)
-        # harvest the oldest reply if it is already here (keeps params fresh)
-        while self.inflight:
-            i = self.inflight[0]
-            s = self.slots[i]
-            if all(r.is_completed() for r in s["reqs"]):
+    def _harvest_replies(self, metrics, block_one=False):
+        """Consume completed replies in order; optionally block for one.
+        A reply superseded by an already-arrived newer one is retired without
+        copying (only the freshest parameters matter)."""
+        blocked = False
+        while self.harvested < self.sent:
+            slot = self.rslots[self.harvested % self.window]
+            if slot["reqs"] is None:
+                break
+            if _done(slot["reqs"]) or (block_one and not blocked):
                 with metrics.timer("comm_wait"):
-                    for r in s["reqs"]:
-                        r.wait()
-                self._worker_apply_params(s, metrics)
-                s["reqs"] = None
-                self.inflight.pop(0)
+                    _wait(slot["reqs"])
+                nxt = self.rslots[(self.harvested + 1) % self.window] \
+                    if self.harvested + 1 < self.sent else None
+                newer = (nxt is not None and nxt["reqs"] is not None
+                         and _done(nxt["reqs"]))
+                self._apply_reply(slot, metrics, skip_copy=newer)
+                blocked = True
             else:
                 break
-        # staleness bound: block if our params are too old
-        if self.worker_step - self.last_applied_step > self.max_stale \
-                and self.inflight:
-            i = self.inflight.pop(0)
-            s = self.slots[i]
+
+    def worker_step_exchange(self, metrics):
+        flat, codec = self.flat, self.codec
+        self.worker_step += 1
+        psh = self.pushes[self.sent % self.window]
+        rsl = self.rslots[self.sent % self.window]
+        if psh["reqs"] is not None:
+            # window full: previous push in this slot must be fully sent and
+            # its reply harvested before the buffers are reused
             with metrics.timer("comm_wait"):
-                for r in s["reqs"]:
-                    r.wait()
-            self._worker_apply_params(s, metrics)
-            s["reqs"] = None
+                _wait(psh["reqs"])
+            if rsl["reqs"] is not None:
+                with metrics.timer("comm_wait"):
+                    _wait(rsl["reqs"])
+                self._apply_reply(rsl, metrics)
+            psh["reqs"] = None
+        with metrics.timer("code_wait"):
+            if codec.name == "identity":
+                psh["wire"].copy_(flat.flat_grad)
+            else:
+                codec.encode(flat.flat_grad, psh["wire"])
+        psh["hdr"][0] = self.worker_step
+        psh["hdr"][1] = self.param_version
+        with metrics.timer("isend_time"):
+            psh["reqs"] = [
+                dist.isend(psh["hdr"], dst=self.comm.ps_rank,
+                           group=self.push_g),
+                dist.isend(psh["wire"], dst=self.comm.ps_rank,
+                           group=self.push_g),
+            ]
+            rsl["reqs"] = [
+                dist.irecv(rsl["hdr"], src=self.comm.ps_rank,
+                           group=self.reply_g),
+                dist.irecv(rsl["buf"], src=self.comm.ps_rank,
+                           group=self.reply_g),
+            ]
+        self.sent += 1
+        # harvest whatever replies already arrived (keeps params fresh)
+        self._harvest_replies(metrics)
+        # bounded staleness: block for one reply if params are too old
+        if self.worker_step - self.last_applied_step > self.max_stale:
+            self._harvest_replies(metrics, block_one=True)
         metrics.add("msg_bytes",
                     self.wire_numel * self.wire_dtype.itemsize
                     + flat.total * flat.dtype.itemsize + 24)
@@ -270,12 +314,12 @@ class AsyncPSEngine:
             if self.dedicated:
                 raise RuntimeError(
                     "dedicated PS rank must call serve(), not step()")
-            # colocated: apply own gradient locally, then serve arrivals
             flat = self.flat
             beta = 1.0 if self._accum_count > 0 else 0.0
-            for b in flat.buckets:
-                ops.reduce_accum(flat.agg_view(b), [flat.grad_view(b)],
-                                 scale=self.gscale, beta=beta)
+            with metrics.timer("decode_time"):
+                for b in flat.buckets:
+                    ops.reduce_accum(flat.agg_view(b), [flat.grad_view(b)],
+                                     scale=self.gscale, beta=beta)
             self._accum_count += 1
             if self._accum_count >= self.quorum:
                 with metrics.timer("optim_step_time"):
@@ -284,6 +328,8 @@ class AsyncPSEngine:
                 self._accum_count = 0
                 self.ps_version += 1
             self.poll_serve(metrics)
+            if self.staleness_hist:
+                metrics["pushes_served"] = sum(self.staleness_hist.values())
         else:
             self.worker_step_exchange(metrics)
 
@@ -296,25 +342,23 @@ class AsyncPSEngine:
         self._apply_fn = self._apply_fn or (lambda b: None)
         if comm.is_ps:
             self.serve(metrics)
-            # wait out the last replies
             for st in self.peers.values():
-                for r in st.reply_reqs:
-                    r.wait()
+                for rep in st.replies:
+                    _wait(rep["reqs"])
         else:
-            # drain outstanding replies
-            while self.inflight:
-                i = self.inflight.pop(0)
-                s = self.slots[i]
-                for r in s["reqs"]:
-                    r.wait()
-                self._worker_apply_params(s, metrics)
-                s["reqs"] = None
-            # send stop
-            dev = self.device
-            hdr = torch.full((2,), -1, dtype=torch.int64, device=dev)
+            # every push gets a reply: drain them all
+            while self.harvested < self.sent:
+                self._harvest_replies(metrics, block_one=True)
+            for psh in self.pushes:
+                if psh["reqs"] is not None:
+                    _wait(psh["reqs"])
+                    psh["reqs"] = None
+            # one stop marker per PS recv slot so every posted irecv matches
+            hdr = torch.full((2,), -1, dtype=torch.int64, device=self.device)
             dummy = torch.zeros(self.wire_numel, dtype=self.wire_dtype,
-                                device=dev)
-            dist.isend(hdr, dst=comm.ps_rank, group=self.group).wait()
-            dist.isend(dummy, dst=comm.ps_rank, group=self.group).wait()
+                                device=self.device)
+            for _ in range(self.ring):
+                dist.isend(hdr, dst=comm.ps_rank, group=self.push_g).wait()
+                dist.isend(dummy, dst=comm.ps_rank, group=self.push_g).wait()
         if comm.initialized:
             dist.barrier()

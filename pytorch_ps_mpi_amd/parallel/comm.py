@@ -64,20 +64,29 @@ class Comm:
         self.rank = dist.get_rank() if self.initialized else 0
         self.world = dist.get_world_size() if self.initialized else 1
         self.ps_rank = ps_rank
-        self.pair_groups = {}
+        # two groups per (PS, worker) pair: the grad-push channel and the
+        # param-reply channel.  With RCCL each group owns its own internal
+        # stream, so a worker's next push never queues behind a pending
+        # reply recv (p2p ops within one communicator are stream-ordered).
+        self.push_groups = {}
+        self.reply_groups = {}
         if make_pair_groups and self.initialized:
             # every rank must create every group, in the same order
             for w in range(self.world):
                 if w == ps_rank:
                     continue
-                self.pair_groups[w] = dist.new_group([ps_rank, w])
+                self.push_groups[w] = dist.new_group([ps_rank, w])
+                self.reply_groups[w] = dist.new_group([ps_rank, w])
 
     @property
     def is_ps(self):
         return self.rank == self.ps_rank
 
-    def pair_group(self, worker_rank):
-        return self.pair_groups[worker_rank]
+    def push_group(self, worker_rank):
+        return self.push_groups[worker_rank]
+
+    def reply_group(self, worker_rank):
+        return self.reply_groups[worker_rank]
 
     def barrier(self):
         if self.initialized:
