@@ -191,3 +191,6 @@ def topk_scatter(dst, idx, val, k, gscale=1.0):
         _EXT.topk_scatter(dst, idx, val, int(k), gscale)
         return
     dst.index_add_(0, idx[:k].long(), val[:k].float() * gscale)
+
+
+from . import bn  # noqa: E402,F401  (fused BatchNorm module; needs ops ready)

@@ -28,6 +28,32 @@ int ps_topk_encode(void* stream, const void* src, int src_is_bf16, int64_t n,
                    int64_t k, uint32_t* ws, int32_t* out_idx, void* out_val);
 int ps_topk_scatter(void* stream, float* dst, const int32_t* idx,
                     const void* val, int val_is_bf16, int64_t k, float gscale);
+int ps_bn_fwd_stats(void* stream, const void* x, float* psum, float* psumsq,
+                    int64_t rows, int64_t C);
+int ps_bn_finalize(void* stream, const float* psum, const float* psumsq,
+                   const void* gamma, const void* beta, void* rmean,
+                   void* rvar, float* mean, float* invstd, float* scale,
+                   float* shift, int64_t C, double count, float eps,
+                   float momentum, int t_is_bf16);
+int ps_bn_eval_coef(void* stream, const void* gamma, const void* beta,
+                    const void* rmean, const void* rvar, float* scale,
+                    float* shift, int64_t C, float eps, int t_is_bf16);
+int ps_bn_normalize(void* stream, const void* x, const void* z, void* y,
+                    const float* scale, const float* shift, int64_t rows,
+                    int64_t C, int relu);
+int ps_bn_bwd_stats(void* stream, const void* x, const void* dy,
+                    const void* z, const float* scale, const float* shift,
+                    float* dsum, float* dxsum, int64_t rows, int64_t C,
+                    int relu);
+int ps_bn_bwd_coef(void* stream, const float* dsum, const float* dxsum,
+                   const float* mean, const float* invstd, const void* gamma,
+                   void* dgamma, void* dbeta, float* ca, float* cbx,
+                   float* cc, int64_t C, double count, int train,
+                   int t_is_bf16);
+int ps_bn_bwd_dx(void* stream, const void* x, const void* dy, const void* z,
+                 void* dx, void* dz, const float* ca, const float* cbx,
+                 const float* cc, const float* scale, const float* shift,
+                 int64_t rows, int64_t C, int relu);
 }
 
 namespace {
@@ -222,9 +248,124 @@ void topk_scatter(at::Tensor dst, at::Tensor idx, at::Tensor val, int64_t k,
            "topk_scatter");
 }
 
+// ---- fused NHWC BatchNorm ------------------------------------------------
+
+void check_bn_stream(const at::Tensor& t, int64_t numel, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == at::kBFloat16, name);
+  TORCH_CHECK(t.numel() == numel, name, " numel mismatch");
+}
+
+const void* opt_ptr(const c10::optional<at::Tensor>& t) {
+  return t.has_value() ? t->data_ptr() : nullptr;
+}
+
+void bn_fwd_stats(at::Tensor x, at::Tensor psum, at::Tensor psumsq,
+                  int64_t rows, int64_t C) {
+  TORCH_CHECK(C % 64 == 0, "C must be a multiple of 64");
+  check_bn_stream(x, rows * C, "x");
+  throw_on(ps_bn_fwd_stats(cur_stream(x), x.data_ptr(),
+                           psum.data_ptr<float>(), psumsq.data_ptr<float>(),
+                           rows, C),
+           "bn_fwd_stats");
+}
+
+void bn_finalize(at::Tensor psum, at::Tensor psumsq,
+                 c10::optional<at::Tensor> gamma,
+                 c10::optional<at::Tensor> beta,
+                 c10::optional<at::Tensor> rmean,
+                 c10::optional<at::Tensor> rvar, at::Tensor mean,
+                 at::Tensor invstd, at::Tensor scale, at::Tensor shift,
+                 double count, double eps, double momentum) {
+  const int64_t C = psum.numel();
+  int bf = gamma.has_value() ? is_bf16(*gamma)
+                             : (rmean.has_value() ? is_bf16(*rmean) : 1);
+  throw_on(ps_bn_finalize(cur_stream(psum), psum.data_ptr<float>(),
+                          psumsq.data_ptr<float>(), opt_ptr(gamma),
+                          opt_ptr(beta), (void*)opt_ptr(rmean),
+                          (void*)opt_ptr(rvar), mean.data_ptr<float>(),
+                          invstd.data_ptr<float>(), scale.data_ptr<float>(),
+                          shift.data_ptr<float>(), C, count, (float)eps,
+                          (float)momentum, bf),
+           "bn_finalize");
+}
+
+void bn_eval_coef(c10::optional<at::Tensor> gamma,
+                  c10::optional<at::Tensor> beta, at::Tensor rmean,
+                  at::Tensor rvar, at::Tensor scale, at::Tensor shift,
+                  double eps) {
+  const int64_t C = rmean.numel();
+  throw_on(ps_bn_eval_coef(cur_stream(rmean), opt_ptr(gamma), opt_ptr(beta),
+                           rmean.data_ptr(), rvar.data_ptr(),
+                           scale.data_ptr<float>(), shift.data_ptr<float>(),
+                           C, (float)eps, is_bf16(rmean)),
+           "bn_eval_coef");
+}
+
+void bn_normalize(at::Tensor x, c10::optional<at::Tensor> z, at::Tensor y,
+                  at::Tensor scale, at::Tensor shift, int64_t rows, int64_t C,
+                  bool relu) {
+  check_bn_stream(x, rows * C, "x");
+  check_bn_stream(y, rows * C, "y");
+  if (z.has_value()) check_bn_stream(*z, rows * C, "z");
+  throw_on(ps_bn_normalize(cur_stream(x), x.data_ptr(), opt_ptr(z),
+                           y.data_ptr(), scale.data_ptr<float>(),
+                           shift.data_ptr<float>(), rows, C, relu ? 1 : 0),
+           "bn_normalize");
+}
+
+void bn_bwd_stats(at::Tensor x, at::Tensor dy, c10::optional<at::Tensor> z,
+                  at::Tensor scale, at::Tensor shift, at::Tensor dsum,
+                  at::Tensor dxsum, int64_t rows, int64_t C, bool relu) {
+  check_bn_stream(x, rows * C, "x");
+  check_bn_stream(dy, rows * C, "dy");
+  throw_on(ps_bn_bwd_stats(cur_stream(x), x.data_ptr(), dy.data_ptr(),
+                           opt_ptr(z), scale.data_ptr<float>(),
+                           shift.data_ptr<float>(), dsum.data_ptr<float>(),
+                           dxsum.data_ptr<float>(), rows, C, relu ? 1 : 0),
+           "bn_bwd_stats");
+}
+
+void bn_bwd_coef(at::Tensor dsum, at::Tensor dxsum, at::Tensor mean,
+                 at::Tensor invstd, c10::optional<at::Tensor> gamma,
+                 c10::optional<at::Tensor> dgamma,
+                 c10::optional<at::Tensor> dbeta, at::Tensor ca,
+                 at::Tensor cbx, at::Tensor cc, double count, bool train) {
+  const int64_t C = dsum.numel();
+  int bf = gamma.has_value() ? is_bf16(*gamma) : 1;
+  throw_on(ps_bn_bwd_coef(cur_stream(dsum), dsum.data_ptr<float>(),
+                          dxsum.data_ptr<float>(), mean.data_ptr<float>(),
+                          invstd.data_ptr<float>(), opt_ptr(gamma),
+                          (void*)opt_ptr(dgamma), (void*)opt_ptr(dbeta),
+                          ca.data_ptr<float>(), cbx.data_ptr<float>(),
+                          cc.data_ptr<float>(), C, count, train ? 1 : 0, bf),
+           "bn_bwd_coef");
+}
+
+void bn_bwd_dx(at::Tensor x, at::Tensor dy, c10::optional<at::Tensor> z,
+               at::Tensor dx, c10::optional<at::Tensor> dz, at::Tensor ca,
+               at::Tensor cbx, at::Tensor cc, at::Tensor scale,
+               at::Tensor shift, int64_t rows, int64_t C, bool relu) {
+  check_bn_stream(x, rows * C, "x");
+  check_bn_stream(dx, rows * C, "dx");
+  TORCH_CHECK(z.has_value() == dz.has_value(), "z/dz pairing");
+  throw_on(ps_bn_bwd_dx(cur_stream(x), x.data_ptr(), dy.data_ptr(),
+                        opt_ptr(z), dx.data_ptr(), (void*)opt_ptr(dz),
+                        ca.data_ptr<float>(), cbx.data_ptr<float>(),
+                        cc.data_ptr<float>(), scale.data_ptr<float>(),
+                        shift.data_ptr<float>(), rows, C, relu ? 1 : 0),
+           "bn_bwd_dx");
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("bn_fwd_stats", &bn_fwd_stats);
+  m.def("bn_finalize", &bn_finalize);
+  m.def("bn_eval_coef", &bn_eval_coef);
+  m.def("bn_normalize", &bn_normalize);
+  m.def("bn_bwd_stats", &bn_bwd_stats);
+  m.def("bn_bwd_coef", &bn_bwd_coef);
+  m.def("bn_bwd_dx", &bn_bwd_dx);
   m.def("fused_sgd", &fused_sgd, "fused SGD over flat fp32 buffers");
   m.def("fused_adam", &fused_adam, "fused Adam over flat fp32 buffers");
   m.def("reduce_accum", &reduce_accum, "dst = beta*dst + scale*sum(srcs)");

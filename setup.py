@@ -23,6 +23,7 @@ setup(
             sources=[
                 os.path.join(CSRC, "bindings.cpp"),
                 os.path.join(CSRC, "ps_kernels.hip"),
+                os.path.join(CSRC, "bn_kernels.hip"),
             ],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
