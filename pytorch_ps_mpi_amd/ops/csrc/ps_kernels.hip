@@ -282,16 +282,30 @@ __device__ __forceinline__ uint32_t tk_key(float x) {
   return cv.u >> 21;  // 11 bits: exponent(8) + mantissa top 3
 }
 
+// Wave-aggregated histogram: gradient magnitudes concentrate in a few dozen
+// key bins, so per-element LDS atomics serialize on hot bins.  Instead each
+// wave resolves its 64 keys into one LDS atomic PER DISTINCT KEY via a
+// ballot loop (iterations = distinct keys in the wave).
 template <typename T>
 __global__ void __launch_bounds__(PS_BLOCK)
 k_topk_hist(const T* __restrict__ src, uint32_t* __restrict__ hist, int64_t n) {
   __shared__ uint32_t lh[TK_BINS];
   for (int b = threadIdx.x; b < TK_BINS; b += blockDim.x) lh[b] = 0;
   __syncthreads();
+  const int lane = threadIdx.x & 63;
   int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = i0; i < n; i += stride) {
-    atomicAdd(&lh[tk_key(ld_as_float(src, i))], 1u);
+    const uint32_t key = tk_key(ld_as_float(src, i));
+    unsigned long long todo = __ballot(1);  // active lanes
+    while (todo) {
+      const int leader = __ffsll(todo) - 1;
+      const uint32_t lkey = __shfl(key, leader, 64);
+      const unsigned long long same = __ballot(key == lkey);
+      if (lane == leader)
+        atomicAdd(&lh[lkey], (uint32_t)__popcll(same & todo));
+      todo &= ~same;
+    }
   }
   __syncthreads();
   for (int b = threadIdx.x; b < TK_BINS; b += blockDim.x) {

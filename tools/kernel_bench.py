@@ -96,6 +96,31 @@ def main():
     t = timeit(lambda: ops.f32_to_bf16(p, bout), args.iters)
     rec("f32_to_bf16", t, n * 6)
 
+    # fused BN vs torch native BN (channels_last bf16), ResNet-50 hot shape
+    import torch.nn as nn
+    from pytorch_ps_mpi_amd.ops.bn import FusedBatchNorm2d
+    N, C, HW = 256, 256, 56
+    cl = torch.channels_last
+    xb = torch.randn(N, C, HW, HW, device=dev).bfloat16().contiguous(
+        memory_format=cl)
+    gy = torch.randn_like(xb).contiguous(memory_format=cl)
+    el = N * C * HW * HW
+    for name, mod in [
+        ("fused_bn", FusedBatchNorm2d(C, relu=True).to(dev, torch.bfloat16)),
+        ("torch_bn", nn.Sequential(nn.BatchNorm2d(C), nn.ReLU()).to(
+            dev, torch.bfloat16)),
+    ]:
+        mod.train()
+
+        def fwdbwd():
+            x = xb.detach().requires_grad_(True)
+            y = mod(x)
+            y.backward(gy)
+
+        t = timeit(fwdbwd, max(5, args.iters // 5))
+        # fwd: r x, w y; bwd: B1 r x,dy; B3 r x,dy w dx  => 2B*(2+2+3+... )
+        rec(f"{name}+relu fwd+bwd", t, el * 2 * 8)
+
     with open("gpurun_out/kernel_bench.json", "w") as f:
         json.dump(results, f, indent=1)
 
