@@ -26,7 +26,7 @@ except Exception as e:  # pragma: no cover - exercised only without built ext
 HAVE_EXT = _EXT is not None
 
 QCHUNK = 256
-TOPK_WS_WORDS = 2048 + 3 + 2
+TOPK_WS_WORDS = 2048 + 4 + 2 * 2048
 
 
 def _require_ext(t: torch.Tensor):

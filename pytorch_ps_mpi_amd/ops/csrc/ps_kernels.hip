@@ -333,51 +333,111 @@ k_topk_plan(const uint32_t* __restrict__ hist, uint32_t* __restrict__ plan, int6
   }
 }
 
-// counters[0]=above slots used, counters[1]=threshold-bin slots used.
-// Slot allocation is wave-aggregated (one global atomic per wave per class,
-// lanes get base + prefix-popcount) — a per-element atomic on one counter
-// saturates at ~88 atomics/us on this chip and was 100x slower.
+// Compaction is 3-phase with NO contended global atomics (a single global
+// counter word takes ~88 atomics/us on this chip; at 1% density nearly every
+// wave carries a candidate, so even wave-aggregated slot allocation cost
+// ~8 ms on 100M elements):
+//   count:  block b owns the contiguous range [b*chunk,(b+1)*chunk) and
+//           counts its above/eq candidates (ballot popcounts, LDS reduce)
+//   scan:   one block turns per-block counts into exclusive offsets
+//   emit:   block b re-reads its range and writes candidates at
+//           offset[b] + block-local LDS-allocated slots (LDS atomics only)
+
+#define TK_NB 2048  // compaction blocks (= per-class offset array length)
+
 template <typename T>
 __global__ void __launch_bounds__(PS_BLOCK)
-k_topk_compact(const T* __restrict__ src, const uint32_t* __restrict__ plan,
-               uint32_t* __restrict__ counters, int32_t* __restrict__ out_idx,
-               T* __restrict__ out_val, int64_t n, int64_t k) {
+k_topk_count(const T* __restrict__ src, const uint32_t* __restrict__ plan,
+             uint32_t* __restrict__ cnt_above, uint32_t* __restrict__ cnt_eq,
+             int64_t n, int64_t chunk) {
+  __shared__ uint32_t red[2][4];
+  const uint32_t thr = plan[0];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int64_t lo = (int64_t)blockIdx.x * chunk;
+  const int64_t hi = min(n, lo + chunk);
+  uint32_t ca = 0, ce = 0;
+  for (int64_t i = lo + threadIdx.x; i < hi; i += PS_BLOCK) {
+    const uint32_t key = tk_key(ld_as_float(src, i));
+    const unsigned long long mab = __ballot(key > thr);
+    const unsigned long long meq = __ballot(key == thr);
+    if (lane == 0) {
+      ca += (uint32_t)__popcll(mab);
+      ce += (uint32_t)__popcll(meq);
+    }
+  }
+  if (lane == 0) { red[0][wave] = ca; red[1][wave] = ce; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    cnt_above[blockIdx.x] = red[0][0] + red[0][1] + red[0][2] + red[0][3];
+    cnt_eq[blockIdx.x] = red[1][0] + red[1][1] + red[1][2] + red[1][3];
+  }
+}
+
+// in-place exclusive scan of both count arrays (nb <= TK_NB, single block)
+__global__ void __launch_bounds__(PS_BLOCK)
+k_topk_scan(uint32_t* __restrict__ cnt_above, uint32_t* __restrict__ cnt_eq,
+            int nb) {
+  if (threadIdx.x == 0) {
+    uint32_t sa = 0, se = 0;
+    for (int b = 0; b < nb; ++b) {
+      const uint32_t a = cnt_above[b], e = cnt_eq[b];
+      cnt_above[b] = sa;
+      cnt_eq[b] = se;
+      sa += a;
+      se += e;
+    }
+  }
+}
+
+template <typename T>
+__global__ void __launch_bounds__(PS_BLOCK)
+k_topk_emit(const T* __restrict__ src, const uint32_t* __restrict__ plan,
+            const uint32_t* __restrict__ off_above,
+            const uint32_t* __restrict__ off_eq,
+            int32_t* __restrict__ out_idx, T* __restrict__ out_val,
+            int64_t n, int64_t chunk) {
+  __shared__ uint32_t l_above, l_eq;
   const uint32_t thr = plan[0];
   const uint32_t n_above = plan[1];
   const uint32_t need = plan[2];
   const int lane = threadIdx.x & 63;
   const unsigned long long lane_lt = (1ull << lane) - 1ull;
-  int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = i0; i < n; i += stride) {
+  if (threadIdx.x == 0) { l_above = 0; l_eq = 0; }
+  __syncthreads();
+  const uint32_t base_above = off_above[blockIdx.x];
+  const uint32_t base_eq = off_eq[blockIdx.x];
+  const int64_t lo = (int64_t)blockIdx.x * chunk;
+  const int64_t hi = min(n, lo + chunk);
+  for (int64_t i = lo + threadIdx.x; i < hi; i += PS_BLOCK) {
     const T raw = src[i];
     const uint32_t key = tk_key(ld_as_float(src, i));
     const bool above = key > thr;
-    const bool eqb = (key == thr) && (need > 0);
+    const bool eqb = (key == thr);
     const unsigned long long mab = __ballot(above);
     const unsigned long long meq = __ballot(eqb);
     if (mab) {
-      uint32_t base = 0;
-      const int leader = __ffsll((unsigned long long)mab) - 1;
-      if (lane == leader)
-        base = atomicAdd(&counters[0], (uint32_t)__popcll(mab));
-      base = __shfl(base, leader, 64);
+      uint32_t wbase = 0;
+      const int leader = __ffsll(mab) - 1;
+      if (lane == leader) wbase = atomicAdd(&l_above, (uint32_t)__popcll(mab));
+      wbase = __shfl(wbase, leader, 64);
       if (above) {
-        const uint32_t slot = base + (uint32_t)__popcll(mab & lane_lt);
-        if (slot < n_above) {  // always true; guard for safety
+        const uint32_t slot = base_above + wbase
+                            + (uint32_t)__popcll(mab & lane_lt);
+        if (slot < n_above) {
           out_idx[slot] = (int32_t)i;
           out_val[slot] = raw;
         }
       }
     }
-    if (meq) {
-      uint32_t base = 0;
-      const int leader = __ffsll((unsigned long long)meq) - 1;
-      if (lane == leader)
-        base = atomicAdd(&counters[1], (uint32_t)__popcll(meq));
-      base = __shfl(base, leader, 64);
+    if (meq && need > 0) {
+      uint32_t wbase = 0;
+      const int leader = __ffsll(meq) - 1;
+      if (lane == leader) wbase = atomicAdd(&l_eq, (uint32_t)__popcll(meq));
+      wbase = __shfl(wbase, leader, 64);
       if (eqb) {
-        const uint32_t eq = base + (uint32_t)__popcll(meq & lane_lt);
+        const uint32_t eq = base_eq + wbase
+                          + (uint32_t)__popcll(meq & lane_lt);
         if (eq < need) {
           const uint32_t slot = n_above + eq;
           out_idx[slot] = (int32_t)i;
@@ -535,33 +595,39 @@ int ps_quant8_reduce(void* stream_, float* dst, const void** scales,
   return (int)hipGetLastError();
 }
 
-// workspace layout (uint32): [hist TK_BINS | plan 3 | counters 2]
-int ps_topk_workspace_words(void) { return TK_BINS + 3 + 2; }
+// workspace layout (uint32): [hist TK_BINS | plan 3 | pad 1 | cntA TK_NB | cntE TK_NB]
+int ps_topk_workspace_words(void) { return TK_BINS + 4 + 2 * TK_NB; }
 
 int ps_topk_encode(void* stream_, const void* src, int src_is_bf16, int64_t n,
                    int64_t k, uint32_t* ws, int32_t* out_idx, void* out_val) {
   hipStream_t stream = (hipStream_t)stream_;
   uint32_t* hist = ws;
   uint32_t* plan = ws + TK_BINS;
-  uint32_t* counters = ws + TK_BINS + 3;
-  hipError_t e = hipMemsetAsync(ws, 0, sizeof(uint32_t) * (TK_BINS + 3 + 2), stream);
+  uint32_t* cnt_a = ws + TK_BINS + 4;
+  uint32_t* cnt_e = cnt_a + TK_NB;
+  hipError_t e = hipMemsetAsync(ws, 0, sizeof(uint32_t) * (TK_BINS + 4), stream);
   if (e != hipSuccess) return (int)e;
-  dim3 grid(ps_grid(n)), block(PS_BLOCK);
-  if (src_is_bf16) {
-    hipLaunchKernelGGL(k_topk_hist<__hip_bfloat16>, grid, block, 0, stream,
-                       (const __hip_bfloat16*)src, hist, n);
-    hipLaunchKernelGGL(k_topk_plan, dim3(1), block, 0, stream, hist, plan, k);
-    hipLaunchKernelGGL(k_topk_compact<__hip_bfloat16>, grid, block, 0, stream,
-                       (const __hip_bfloat16*)src, plan, counters, out_idx,
-                       (__hip_bfloat16*)out_val, n, k);
-  } else {
-    hipLaunchKernelGGL(k_topk_hist<float>, grid, block, 0, stream,
-                       (const float*)src, hist, n);
-    hipLaunchKernelGGL(k_topk_plan, dim3(1), block, 0, stream, hist, plan, k);
-    hipLaunchKernelGGL(k_topk_compact<float>, grid, block, 0, stream,
-                       (const float*)src, plan, counters, out_idx,
-                       (float*)out_val, n, k);
-  }
+  // contiguous block partition for deterministic per-block offsets
+  int nb = TK_NB;
+  int64_t chunk = (n + nb - 1) / nb;
+  chunk = (chunk + PS_BLOCK - 1) / PS_BLOCK * PS_BLOCK;
+  nb = (int)((n + chunk - 1) / chunk);
+  dim3 grid(ps_grid(n)), cgrid(nb), block(PS_BLOCK);
+#define TK_RUN(T)                                                              \
+  do {                                                                         \
+    hipLaunchKernelGGL(k_topk_hist<T>, grid, block, 0, stream,                 \
+                       (const T*)src, hist, n);                                \
+    hipLaunchKernelGGL(k_topk_plan, dim3(1), block, 0, stream, hist, plan, k); \
+    hipLaunchKernelGGL(k_topk_count<T>, cgrid, block, 0, stream,               \
+                       (const T*)src, plan, cnt_a, cnt_e, n, chunk);           \
+    hipLaunchKernelGGL(k_topk_scan, dim3(1), block, 0, stream, cnt_a, cnt_e,   \
+                       nb);                                                    \
+    hipLaunchKernelGGL(k_topk_emit<T>, cgrid, block, 0, stream, (const T*)src, \
+                       plan, cnt_a, cnt_e, out_idx, (T*)out_val, n, chunk);    \
+  } while (0)
+  if (src_is_bf16) TK_RUN(__hip_bfloat16);
+  else TK_RUN(float);
+#undef TK_RUN
   return (int)hipGetLastError();
 }
 
