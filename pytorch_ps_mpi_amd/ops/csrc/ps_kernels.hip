@@ -374,18 +374,44 @@ k_topk_count(const T* __restrict__ src, const uint32_t* __restrict__ plan,
   }
 }
 
-// in-place exclusive scan of both count arrays (nb <= TK_NB, single block)
+// in-place exclusive scan of both count arrays (nb <= TK_NB, single block).
+// Parallel: each thread serial-scans an 8-entry strip in registers; the 256
+// strip totals are scanned in LDS; strip offsets added back.  (A serial
+// thread-0 loop over global memory was ~0.4 ms of dependent latency.)
 __global__ void __launch_bounds__(PS_BLOCK)
 k_topk_scan(uint32_t* __restrict__ cnt_above, uint32_t* __restrict__ cnt_eq,
             int nb) {
-  if (threadIdx.x == 0) {
-    uint32_t sa = 0, se = 0;
-    for (int b = 0; b < nb; ++b) {
-      const uint32_t a = cnt_above[b], e = cnt_eq[b];
-      cnt_above[b] = sa;
-      cnt_eq[b] = se;
-      sa += a;
-      se += e;
+  __shared__ uint32_t part[2][PS_BLOCK + 1];
+  const int t = threadIdx.x;
+  const int strip = (TK_NB / PS_BLOCK);  // 8
+  uint32_t va[8], ve[8];
+  uint32_t sa = 0, se = 0;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int b = t * strip + j;
+    const uint32_t a = (b < nb) ? cnt_above[b] : 0;
+    const uint32_t e = (b < nb) ? cnt_eq[b] : 0;
+    va[j] = sa; ve[j] = se;  // local exclusive prefix
+    sa += a; se += e;
+  }
+  part[0][t + 1] = sa;
+  part[1][t + 1] = se;
+  if (t == 0) { part[0][0] = 0; part[1][0] = 0; }
+  __syncthreads();
+  if (t == 0) {  // 256 serial LDS adds (~10 us worst case)
+    for (int i = 1; i <= PS_BLOCK; ++i) {
+      part[0][i] += part[0][i - 1];
+      part[1][i] += part[1][i - 1];
+    }
+  }
+  __syncthreads();
+  const uint32_t oa = part[0][t], oe = part[1][t];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int b = t * strip + j;
+    if (b < nb) {
+      cnt_above[b] = oa + va[j];
+      cnt_eq[b] = oe + ve[j];
     }
   }
 }
