@@ -36,7 +36,8 @@ from .utils.metrics import StepMetrics
 class PS(torch.optim.Optimizer):
     def __init__(self, named_params, defaults, *, code=None, mode="replicated",
                  bucket_mb=50, grad_scale="sum", window=2, max_stale=8,
-                 quorum=1, dedicated_ps=False, dtype=None):
+                 quorum=1, dedicated_ps=False, dtype=None, overlap=True,
+                 debug_consistency=0):
         named_params = list(named_params)
         if named_params and not isinstance(named_params[0], tuple):
             raise TypeError("pass model.named_parameters(), not parameters()")
@@ -83,9 +84,23 @@ class PS(torch.optim.Optimizer):
             dist.broadcast(self.flat.flat_param, src=0)
             self.flat.sync_master_from_param()
 
+        # backward-hook comm overlap (replaces the reference's 200-thread
+        # encode pool, ps.py:85,98-101): each bucket's collective launches
+        # from autograd as soon as its grads are complete.
+        self._hook_handles = []
+        if overlap and getattr(self.engine, "wants_hooks", False) \
+                and self.comm.world > 1:
+            for p in params:
+                self._hook_handles.append(
+                    p.register_post_accumulate_grad_hook(
+                        self.engine.on_param_grad))
+
         self._step_count = 0
         self._bucket_apply_count = {}
         self.names = [n for n, p in named_params if p.requires_grad]
+        # debug: cross-rank param-consistency check every N steps (SURVEY §5
+        # race-detection gap; replicated mode must stay bitwise identical)
+        self.debug_consistency = int(debug_consistency)
         self._alloc_state()
 
     # ------------------------------------------------------------------
@@ -110,6 +125,8 @@ class PS(torch.optim.Optimizer):
 
     def zero_grad(self, set_to_none=False):  # noqa: ARG002 (flat buffers)
         self.flat.zero_grad()
+        if self._hook_handles:
+            self.engine.start_step()
 
     def step(self, closure=None, loss=None):
         """Run one exchange+update. Returns (loss, metrics) like ps.py:193."""
@@ -120,6 +137,16 @@ class PS(torch.optim.Optimizer):
         self._step_count += 1
         self.engine.step(self._apply_bucket, metrics)
         metrics["step"] = self._step_count
+        if (self.debug_consistency and self.comm.initialized
+                and self.mode == "replicated"
+                and self._step_count % self.debug_consistency == 0):
+            cs = [None] * self.comm.world
+            dist.all_gather_object(cs, self.flat.param_checksum())
+            metrics["consistent"] = len(set(cs)) == 1
+            if not metrics["consistent"]:
+                raise RuntimeError(
+                    f"replicated params diverged at step {self._step_count}: "
+                    f"checksums {cs}")
         metrics["wire_codec"] = self.codec.name
         if self.mode == "async" and self.comm.is_ps:
             metrics["staleness_hist"] = dict(

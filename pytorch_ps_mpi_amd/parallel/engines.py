@@ -49,7 +49,18 @@ class LocalEngine:
 
 
 class ReplicatedEngine:
+    """Replicated-update DP with backward-hook comm overlap.
+
+    The reference barriered on all encodes before any traffic (ps.py:129);
+    here each bucket's collective launches as soon as its last gradient is
+    accumulated, in FIXED bucket order on every rank (RCCL collectives must
+    be issued in identical order), overlapping comm with the rest of
+    backward.  Buckets whose hooks never fired (unused params / no hooks)
+    are launched at step() time.
+    """
+
     name = "replicated"
+    wants_hooks = True
 
     def __init__(self, flat, codec, comm, grad_scale=1.0):
         self.flat = flat
@@ -57,6 +68,9 @@ class ReplicatedEngine:
         self.comm = comm
         self.gscale = grad_scale
         self.use_allreduce = getattr(codec, "supports_allreduce", False)
+        self._works = {}
+        self._next = 0
+        self._ready = {}
         if not self.use_allreduce:
             dev = flat.flat_param.device
             self.wire_send = {}
@@ -71,44 +85,65 @@ class ReplicatedEngine:
                     for _ in range(comm.world)
                 ]
 
-    def step(self, apply_fn, metrics):
-        flat, codec, comm = self.flat, self.codec, self.comm
-        works = []
+    # ---- hook-driven overlap -----------------------------------------
+
+    def start_step(self):
+        self._works = {}
+        self._next = 0
+        self._ready = {b.idx: 0 for b in self.flat.buckets}
+
+    def on_param_grad(self, param):
+        b = self.flat.param_to_bucket.get(param)
+        if b is None or not self._ready:
+            return
+        self._ready[b.idx] += 1
+        # launch every fully-ready bucket at the head of the fixed schedule
+        while self._next < len(self.flat.buckets):
+            nb = self.flat.buckets[self._next]
+            if self._ready.get(nb.idx, 0) < len(nb.params):
+                break
+            self._launch(nb)
+            self._next += 1
+
+    def _launch(self, b):
+        flat = self.flat
         if self.use_allreduce:
-            with metrics.timer("isend_time"):
-                for b in flat.buckets:
-                    works.append(dist.all_reduce(flat.grad_view(b),
-                                                 async_op=True))
-            nbytes = flat.total * flat.dtype.itemsize
-            for b, w in zip(flat.buckets, works):
-                with metrics.timer("comm_wait"):
-                    w.wait()
-                with metrics.timer("decode_time"):
+            self._works[b.idx] = dist.all_reduce(flat.grad_view(b),
+                                                 async_op=True)
+        else:
+            self.codec.encode(flat.grad_view(b), self.wire_send[b.idx])
+            self._works[b.idx] = dist.all_gather(self.wire_slots[b.idx],
+                                                 self.wire_send[b.idx],
+                                                 async_op=True)
+
+    # ---- step ---------------------------------------------------------
+
+    def step(self, apply_fn, metrics):
+        flat = self.flat
+        with metrics.timer("isend_time"):
+            for b in flat.buckets:  # launch whatever the hooks did not
+                if b.idx not in self._works:
+                    self._launch(b)
+        nbytes = flat.total * flat.dtype.itemsize if self.use_allreduce \
+            else sum(w.numel() * w.dtype.itemsize
+                     for w in self.wire_send.values())
+        for b in flat.buckets:
+            with metrics.timer("comm_wait"):
+                self._works[b.idx].wait()
+            with metrics.timer("decode_time"):
+                if self.use_allreduce:
                     # all-reduce already summed across ranks -> cast+scale
                     ops.reduce_accum(flat.agg_view(b), [flat.grad_view(b)],
                                      scale=self.gscale, beta=0.0)
-                with metrics.timer("optim_step_time"):
-                    apply_fn(b)
-        else:
-            with metrics.timer("code_wait"):
-                for b in flat.buckets:
-                    codec.encode(flat.grad_view(b), self.wire_send[b.idx])
-            with metrics.timer("isend_time"):
-                for b in flat.buckets:
-                    works.append(dist.all_gather(self.wire_slots[b.idx],
-                                                 self.wire_send[b.idx],
-                                                 async_op=True))
-            nbytes = sum(w.numel() * w.dtype.itemsize
-                         for w in self.wire_send.values())
-            for b, w in zip(flat.buckets, works):
-                with metrics.timer("comm_wait"):
-                    w.wait()
-                with metrics.timer("decode_time"):
-                    codec.decode_reduce(flat.agg_view(b),
-                                        self.wire_slots[b.idx],
-                                        gscale=self.gscale, beta=0.0)
-                with metrics.timer("optim_step_time"):
-                    apply_fn(b)
+                else:
+                    self.codec.decode_reduce(flat.agg_view(b),
+                                             self.wire_slots[b.idx],
+                                             gscale=self.gscale, beta=0.0,
+                                             src_dtype=flat.dtype)
+            with metrics.timer("optim_step_time"):
+                apply_fn(b)
+        self._works = {}
+        self._ready = {}
         metrics.add("msg_bytes", nbytes)
 
     def finish(self):

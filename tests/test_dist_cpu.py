@@ -76,7 +76,7 @@ def _replicated_worker(rank, port, codec, out_file):
     model, _full, (xs, ys) = _mlp_and_data(rank)
     opt = SGD(model.named_parameters(), lr=0.05, momentum=0.9,
               mode="replicated", code=codec, grad_scale="mean",
-              bucket_mb=0.05)
+              bucket_mb=0.05, debug_consistency=2)
     for step in range(5):
         opt.zero_grad()
         loss = models.loss_fn("mlp", model, xs, ys)
