@@ -116,10 +116,13 @@ class AsyncPSEngine:
                            self.ring, reply_ring, wn, self.wire_dtype,
                            total, flat.dtype, dev)
                 self.peers[w] = st
-            # pre-post every recv slot, per peer, in ring order
-            for st in self.peers.values():
-                for s in st.slots:
-                    st.post(s)
+            # recv-ring posting is DEFERRED to first step()/serve(): with
+            # RCCL the first op on a pair communicator blocks the host in
+            # ncclCommInitRank until the peer joins, and at __init__ time
+            # the peers are still waiting for the optimizer's initial param
+            # broadcast (deadlock).  gloo initializes groups eagerly and
+            # does not care.
+            self._started = False
         else:
             self.push_g = comm.push_group(comm.rank)
             self.reply_g = comm.reply_group(comm.rank)
@@ -140,6 +143,15 @@ class AsyncPSEngine:
             self.last_applied_step = 0
 
     # ------------------------------------------------------------------ PS
+
+    def _start_ps(self):
+        """Pre-post every recv slot, per peer, in ring order (first use)."""
+        if self._started:
+            return
+        self._started = True
+        for st in self.peers.values():
+            for s in st.slots:
+                st.post(s)
 
     def _serve_slot(self, st, metrics):
         """Process the head recv slot of peer st (must be completed/waited)."""
@@ -196,6 +208,7 @@ class AsyncPSEngine:
 
     def poll_serve(self, metrics, max_per_peer=None):
         """Serve arrived pushes without blocking. PS-side only."""
+        self._start_ps()
         served = 0
         budget = max_per_peer if max_per_peer is not None else self.ring
         for st in self.peers.values():
@@ -210,6 +223,7 @@ class AsyncPSEngine:
 
     def serve(self, metrics):
         """Dedicated-PS loop: serve until every peer sent its stop."""
+        self._start_ps()
         while True:
             alive = [st for st in self.peers.values() if not st.stopped]
             if not alive:
