@@ -180,6 +180,12 @@ def main():
                 "bucket_mb": args.bucket_mb,
             },
         }
+        if args.mode == "async" and world > 1:
+            # BASELINE config 5: staleness distribution (PS-side histogram
+            # of ps_version - version_used_by_worker per served push)
+            hist = dict(getattr(opt.engine, "staleness_hist", {}))
+            out["config"]["staleness_hist"] = \
+                {str(k): v for k, v in sorted(hist.items())}
         print(json.dumps(out))
 
     if dist.is_initialized():

@@ -37,7 +37,7 @@ class PS(torch.optim.Optimizer):
     def __init__(self, named_params, defaults, *, code=None, mode="replicated",
                  bucket_mb=50, grad_scale="sum", window=2, max_stale=8,
                  quorum=1, dedicated_ps=False, dtype=None, overlap=True,
-                 debug_consistency=0):
+                 debug_consistency=0, profile_gpu=False):
         named_params = list(named_params)
         if named_params and not isinstance(named_params[0], tuple):
             raise TypeError("pass model.named_parameters(), not parameters()")
@@ -101,6 +101,7 @@ class PS(torch.optim.Optimizer):
         # debug: cross-rank param-consistency check every N steps (SURVEY §5
         # race-detection gap; replicated mode must stay bitwise identical)
         self.debug_consistency = int(debug_consistency)
+        self.profile_gpu = bool(profile_gpu)
         self._alloc_state()
 
     # ------------------------------------------------------------------
@@ -130,12 +131,13 @@ class PS(torch.optim.Optimizer):
 
     def step(self, closure=None, loss=None):
         """Run one exchange+update. Returns (loss, metrics) like ps.py:193."""
-        metrics = StepMetrics()
+        metrics = StepMetrics(gpu=self.profile_gpu)
         if closure is not None:
             with torch.enable_grad():
                 loss = closure()
         self._step_count += 1
         self.engine.step(self._apply_bucket, metrics)
+        metrics.finalize_gpu()
         metrics["step"] = self._step_count
         if (self.debug_consistency and self.comm.initialized
                 and self.mode == "replicated"

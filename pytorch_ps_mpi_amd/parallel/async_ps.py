@@ -39,9 +39,24 @@ def _done(reqs):
     return all(r.is_completed() for r in reqs)
 
 
-def _wait(reqs):
+def _wait(reqs, timeout_s=None, what=""):
+    """Wait all reqs; with timeout_s, log + raise if a peer goes silent
+    (the reference assumed 'communication is reliable', README.md:7-8 —
+    this is the PS-side timeout logging SURVEY §5 calls for)."""
+    if timeout_s is None:
+        for r in reqs:
+            r.wait()
+        return
+    import datetime
     for r in reqs:
-        r.wait()
+        try:
+            r.wait(datetime.timedelta(seconds=timeout_s))
+        except Exception as e:
+            import logging
+            logging.getLogger(__name__).error(
+                "async-PS wait timed out after %ss (%s): %r",
+                timeout_s, what, e)
+            raise
 
 
 class _Peer:
@@ -80,7 +95,8 @@ class AsyncPSEngine:
     name = "async_ps"
 
     def __init__(self, flat, codec, comm, grad_scale=1.0, window=4,
-                 max_stale=8, quorum=1, dedicated=False, reply_ring=2):
+                 max_stale=8, quorum=1, dedicated=False, reply_ring=2,
+                 serve_timeout_s=None):
         self.flat = flat
         self.codec = codec
         self.comm = comm
@@ -90,6 +106,7 @@ class AsyncPSEngine:
         self.max_stale = int(max_stale)
         self.quorum = max(1, int(quorum))
         self.dedicated = bool(dedicated)
+        self.serve_timeout_s = serve_timeout_s
         self._apply_fn = None
 
         dev = flat.flat_param.device
@@ -153,11 +170,11 @@ class AsyncPSEngine:
             for s in st.slots:
                 st.post(s)
 
-    def _serve_slot(self, st, metrics):
+    def _serve_slot(self, st, metrics, timeout_s=None):
         """Process the head recv slot of peer st (must be completed/waited)."""
         flat, codec = self.flat, self.codec
         slot = st.slots[st.head]
-        _wait(slot["reqs"])
+        _wait(slot["reqs"], timeout_s, f"push from worker {st.rank}")
         slot["reqs"] = None
         hdr = slot["hdr"].tolist()
         if hdr[0] < 0:
@@ -231,7 +248,8 @@ class AsyncPSEngine:
             progressed = self.poll_serve(metrics)
             if progressed == 0:
                 # block on one peer's head slot to make progress
-                self._serve_slot(alive[0], metrics)
+                self._serve_slot(alive[0], metrics,
+                                 timeout_s=self.serve_timeout_s)
 
     # -------------------------------------------------------------- worker
 

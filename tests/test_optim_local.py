@@ -134,3 +134,21 @@ def test_gpt2_tiny_forward_backward():
     loss = m.loss(x[:, :-1], x[:, 1:])
     loss.backward()
     assert torch.isfinite(loss)
+
+
+def test_print_summary(capsys):
+    from pytorch_ps_mpi_amd.utils.metrics import print_summary
+    torch.manual_seed(0)
+    m = models.build_model("mlp")
+    opt = SGD(m.named_parameters(), lr=0.1)
+    ms = []
+    x, y = models.synthetic_batch("mlp", 8, seed=0)
+    for _ in range(3):
+        opt.zero_grad()
+        loss = models.loss_fn("mlp", m, x, y)
+        loss.backward()
+        _, metrics = opt.step(loss=loss)
+        ms.append(metrics)
+    print_summary(ms)
+    outp = capsys.readouterr().out
+    assert "optim_step_time" in outp and "mean" in outp
