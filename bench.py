@@ -146,7 +146,10 @@ def main():
 
     # max over training ranks (dedicated PS contributes 0)
     if dist.is_initialized():
+        if on_gpu:
+            elapsed = elapsed.to(device)  # nccl needs device tensors
         dist.all_reduce(elapsed, op=dist.ReduceOp.MAX)
+        elapsed = elapsed.cpu()
     barrier_sync()
 
     t = float(elapsed.item())
