@@ -192,6 +192,11 @@ k_bn_eval_coef(const T* __restrict__ gamma, const T* __restrict__ beta,
 // lane handles 8 consecutive channels; scale/shift gathered from L2.
 // ---------------------------------------------------------------------------
 
+// NOTE on the grid: the launcher rounds gridDim so that
+// gridDim.x * BN_BLOCK is a multiple of C8 — then each thread's channel
+// group is LOOP-INVARIANT and the per-channel coefficients are loaded once
+// before the loop (per-iteration scalar coefficient gathers made the first
+// version of these kernels VMEM-issue-bound: 44% of a ResNet-50 step).
 template <bool HAS_Z>
 __global__ void __launch_bounds__(BN_BLOCK)
 k_bn_normalize(const __hip_bfloat16* __restrict__ x,
@@ -202,17 +207,17 @@ k_bn_normalize(const __hip_bfloat16* __restrict__ x,
                int64_t n8, int64_t C8, int relu) {
   int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const int64_t cb = (i0 % C8) * 8;  // loop-invariant: stride % C8 == 0
+  float sc[8], sh[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    sc[j] = scale[cb + j];
+    sh[j] = shift[cb + j];
+  }
   for (int64_t i = i0; i < n8; i += stride) {
-    const int64_t cb = (i % C8) * 8;
     const bnx8 xv = ((const bnx8*)x)[i];
     bnx8 zv;
     if (HAS_Z) zv = ((const bnx8*)z)[i];
-    const float4 s01 = ((const float4*)(scale + cb))[0];
-    const float4 s23 = ((const float4*)(scale + cb))[1];
-    const float4 h01 = ((const float4*)(shift + cb))[0];
-    const float4 h23 = ((const float4*)(shift + cb))[1];
-    const float sc[8] = {s01.x, s01.y, s01.z, s01.w, s23.x, s23.y, s23.z, s23.w};
-    const float sh[8] = {h01.x, h01.y, h01.z, h01.w, h23.x, h23.y, h23.z, h23.w};
     bnx8 out;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
@@ -243,20 +248,20 @@ k_bn_bwd_dx(const __hip_bfloat16* __restrict__ x,
             int64_t n8, int64_t C8, int relu) {
   int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const int64_t cb = (i0 % C8) * 8;  // loop-invariant: stride % C8 == 0
+  float av[8], bxv[8], ccv[8], scv[8], shv[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    av[j] = ca[cb + j];
+    bxv[j] = cbx[cb + j];
+    ccv[j] = cc[cb + j];
+    if (relu) { scv[j] = scale[cb + j]; shv[j] = shift[cb + j]; }
+  }
   for (int64_t i = i0; i < n8; i += stride) {
-    const int64_t cb = (i % C8) * 8;
     const bnx8 xv = ((const bnx8*)x)[i];
     const bnx8 dyv = ((const bnx8*)dy)[i];
     bnx8 zv;
     if (HAS_Z) zv = ((const bnx8*)z)[i];
-    float av[8], bxv[8], ccv[8], scv[8], shv[8];
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      av[j] = ca[cb + j];
-      bxv[j] = cbx[cb + j];
-      ccv[j] = cc[cb + j];
-      if (relu) { scv[j] = scale[cb + j]; shv[j] = shift[cb + j]; }
-    }
     bnx8 odx, odz;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
@@ -321,10 +326,19 @@ static inline dim3 bn_reduce_grid(int64_t rows, int64_t C) {
   return dim3(ct, (unsigned)yb);
 }
 
-static inline int bn_elem_grid(int64_t n8) {
+static inline int64_t bn_gcd(int64_t a, int64_t b) {
+  while (b) { int64_t t = a % b; a = b; b = t; }
+  return a;
+}
+
+// grid such that gridDim * BN_BLOCK is a multiple of C8 (loop-invariant
+// channel groups in the elementwise kernels)
+static inline int bn_elem_grid(int64_t n8, int64_t C8) {
   int64_t b = (n8 + BN_BLOCK - 1) / BN_BLOCK;
   if (b > 2048) b = 2048;
   if (b < 1) b = 1;
+  const int64_t f = C8 / bn_gcd(C8, (int64_t)BN_BLOCK);
+  b = (b + f - 1) / f * f;
   return (int)b;
 }
 
@@ -379,7 +393,7 @@ int ps_bn_normalize(void* stream_, const void* x, const void* z, void* y,
                     int64_t C, int relu) {
   hipStream_t s = (hipStream_t)stream_;
   const int64_t n8 = rows * C / 8;
-  dim3 grid(bn_elem_grid(n8));
+  dim3 grid(bn_elem_grid(n8, C / 8));
   if (z != nullptr)
     hipLaunchKernelGGL((k_bn_normalize<true>), grid, dim3(BN_BLOCK), 0, s,
                        (const __hip_bfloat16*)x, (const __hip_bfloat16*)z,
@@ -435,7 +449,7 @@ int ps_bn_bwd_dx(void* stream_, const void* x, const void* dy, const void* z,
                  int64_t rows, int64_t C, int relu) {
   hipStream_t s = (hipStream_t)stream_;
   const int64_t n8 = rows * C / 8;
-  dim3 grid(bn_elem_grid(n8));
+  dim3 grid(bn_elem_grid(n8, C / 8));
   if (z != nullptr)
     hipLaunchKernelGGL((k_bn_bwd_dx<true>), grid, dim3(BN_BLOCK), 0, s,
                        (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy,
