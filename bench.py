@@ -100,6 +100,20 @@ def main():
     if channels_last:
         x = x.contiguous(memory_format=torch.channels_last)
 
+    # MIOpen find warm-up: with N ranks on one box every process would
+    # compile the same conv solutions concurrently (8x contention, many
+    # minutes).  Rank 0 populates the shared find/binary cache with
+    # comm-free forward+backward passes; the rest wait at a barrier.
+    # Safe in async mode: the PS posts nothing before its first step().
+    if on_gpu and world > 1 and dist.is_initialized():
+        if rank == 0:
+            for _ in range(2):
+                opt.zero_grad()
+                models.loss_fn(args.model, model, x, y).backward()
+            opt.zero_grad()
+            torch.cuda.synchronize()
+        dist.barrier()
+
     is_serving_ps = (args.mode == "async" and args.dedicated_ps and
                      world > 1 and rank == 0)
     n_train = world - 1 if (args.mode == "async" and args.dedicated_ps

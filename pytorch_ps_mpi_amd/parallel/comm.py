@@ -24,7 +24,7 @@ def env_world():
     return int(os.environ.get("WORLD_SIZE", "1"))
 
 
-def init_distributed(backend=None, device=None, timeout_s=300):
+def init_distributed(backend=None, device=None, timeout_s=1800):
     """Initialise torch.distributed from torchrun/torch.distributed.run env.
 
     Returns the torch.device this rank should use.  Safe to call when
