@@ -23,11 +23,6 @@ from . import HAVE_EXT, _EXT
 _CL = torch.channels_last
 
 
-def _flatview(t):
-    # NHWC-contiguous [N,C,H,W] tensor: memory is [N*H*W, C]
-    return t
-
-
 class _FusedBN(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, z, weight, bias, rmean, rvar, training, momentum,

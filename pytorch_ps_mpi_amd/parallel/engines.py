@@ -205,14 +205,10 @@ class SyncPSEngine:
                 w.wait()
             if comm.is_ps:
                 with metrics.timer("decode_time"):
-                    if self.ident:
-                        codec.decode_reduce(flat.agg_view(b),
-                                            self.wire_slots[b.idx],
-                                            gscale=self.gscale, beta=0.0)
-                    else:
-                        codec.decode_reduce(flat.agg_view(b),
-                                            self.wire_slots[b.idx],
-                                            gscale=self.gscale, beta=0.0)
+                    codec.decode_reduce(flat.agg_view(b),
+                                        self.wire_slots[b.idx],
+                                        gscale=self.gscale, beta=0.0,
+                                        src_dtype=flat.dtype)
                 with metrics.timer("optim_step_time"):
                     apply_fn(b)
             bworks.append(dist.broadcast(flat.param_view(b), src=comm.ps_rank,

@@ -29,7 +29,7 @@ ALIGN = 256
 
 
 class Bucket:
-    __slots__ = ("idx", "start", "end", "params", "ready", "_views")
+    __slots__ = ("idx", "start", "end", "params", "ready")
 
     def __init__(self, idx, start, end, params):
         self.idx = idx
@@ -37,7 +37,6 @@ class Bucket:
         self.end = end
         self.params = params  # list[(name, param, offset, numel)]
         self.ready = 0
-        self._views = {}
 
     @property
     def numel(self):
