@@ -6,15 +6,17 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops.ln import FusedLayerNorm
+
 
 class Block(nn.Module):
     def __init__(self, dim, heads, mlp_ratio=4.0):
         super().__init__()
-        self.n1 = nn.LayerNorm(dim)
+        self.n1 = FusedLayerNorm(dim)
         self.qkv = nn.Linear(dim, dim * 3)
         self.proj = nn.Linear(dim, dim)
         self.heads = heads
-        self.n2 = nn.LayerNorm(dim)
+        self.n2 = FusedLayerNorm(dim)
         h = int(dim * mlp_ratio)
         self.fc1 = nn.Linear(dim, h)
         self.fc2 = nn.Linear(h, dim)
@@ -34,16 +36,29 @@ class ViT(nn.Module):
     def __init__(self, img=224, patch=16, dim=768, depth=12, heads=12,
                  num_classes=1000):
         super().__init__()
-        self.patch_embed = nn.Conv2d(3, dim, patch, patch)
+        # patch embedding as unfold + Linear: with stride == kernel the
+        # "conv" is a pure reshape + GEMM, which keeps it on hipBLASLt
+        # (MIOpen's stride-16 bf16 conv fell back to naive/im2col kernels —
+        # 60% of the step in the round-1 profile)
+        self.patch = patch
+        self.patch_embed = nn.Linear(3 * patch * patch, dim)
         n = (img // patch) ** 2
         self.cls = nn.Parameter(torch.zeros(1, 1, dim))
         self.pos = nn.Parameter(torch.randn(1, n + 1, dim) * 0.02)
         self.blocks = nn.ModuleList(Block(dim, heads) for _ in range(depth))
-        self.norm = nn.LayerNorm(dim)
+        self.norm = FusedLayerNorm(dim)
         self.head = nn.Linear(dim, num_classes)
 
+    def _patchify(self, x):
+        B, C, H, W = x.shape
+        p = self.patch
+        x = x.view(B, C, H // p, p, W // p, p)
+        x = x.permute(0, 2, 4, 1, 3, 5).reshape(B, (H // p) * (W // p),
+                                                C * p * p)
+        return x
+
     def forward(self, x):
-        x = self.patch_embed(x).flatten(2).transpose(1, 2)
+        x = self.patch_embed(self._patchify(x))
         cls = self.cls.expand(x.shape[0], -1, -1)
         x = torch.cat([cls, x], dim=1) + self.pos
         for blk in self.blocks:

@@ -54,6 +54,15 @@ int ps_bn_bwd_dx(void* stream, const void* x, const void* dy, const void* z,
                  void* dx, void* dz, const float* ca, const float* cbx,
                  const float* cc, const float* scale, const float* shift,
                  int64_t rows, int64_t C, int relu);
+int ps_ln_fwd(void* stream, const void* x, void* y, const void* gamma,
+              const void* beta, float* mean, float* rstd, int64_t rows,
+              int64_t D, float eps);
+int ps_ln_bwd_dx(void* stream, const void* x, const void* dy, void* dx,
+                 const void* gamma, const float* mean, const float* rstd,
+                 int64_t rows, int64_t D);
+int ps_ln_bwd_dgb(void* stream, const void* x, const void* dy,
+                  const float* mean, const float* rstd, float* dgamma,
+                  float* dbeta, int64_t rows, int64_t D);
 }
 
 namespace {
@@ -356,9 +365,44 @@ void bn_bwd_dx(at::Tensor x, at::Tensor dy, c10::optional<at::Tensor> z,
            "bn_bwd_dx");
 }
 
+// ---- fused row-wise LayerNorm -------------------------------------------
+
+void ln_fwd(at::Tensor x, at::Tensor y, at::Tensor gamma, at::Tensor beta,
+            at::Tensor mean, at::Tensor rstd, int64_t rows, int64_t D,
+            double eps) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16, "x");
+  TORCH_CHECK(gamma.scalar_type() == at::kBFloat16, "gamma dtype");
+  throw_on(ps_ln_fwd(cur_stream(x), x.data_ptr(), y.data_ptr(),
+                     gamma.data_ptr(), beta.data_ptr(),
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(), rows, D,
+                     (float)eps),
+           "ln_fwd");
+}
+
+void ln_bwd_dx(at::Tensor x, at::Tensor dy, at::Tensor dx, at::Tensor gamma,
+               at::Tensor mean, at::Tensor rstd, int64_t rows, int64_t D) {
+  throw_on(ps_ln_bwd_dx(cur_stream(x), x.data_ptr(), dy.data_ptr(),
+                        dx.data_ptr(), gamma.data_ptr(),
+                        mean.data_ptr<float>(), rstd.data_ptr<float>(), rows,
+                        D),
+           "ln_bwd_dx");
+}
+
+void ln_bwd_dgb(at::Tensor x, at::Tensor dy, at::Tensor mean, at::Tensor rstd,
+                at::Tensor dgamma, at::Tensor dbeta, int64_t rows, int64_t D) {
+  throw_on(ps_ln_bwd_dgb(cur_stream(x), x.data_ptr(), dy.data_ptr(),
+                         mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                         dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                         rows, D),
+           "ln_bwd_dgb");
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("ln_fwd", &ln_fwd);
+  m.def("ln_bwd_dx", &ln_bwd_dx);
+  m.def("ln_bwd_dgb", &ln_bwd_dgb);
   m.def("bn_fwd_stats", &bn_fwd_stats);
   m.def("bn_finalize", &bn_finalize);
   m.def("bn_eval_coef", &bn_eval_coef);
