@@ -97,6 +97,14 @@ class ReplicatedEngine:
         if b is None or not self._ready:
             return
         self._ready[b.idx] += 1
+        if self._ready[b.idx] > len(b.params):
+            # a second backward before step(): the bucket's collective may
+            # already be in flight with PARTIAL gradients — silent wrong
+            # results.  Gradient accumulation requires overlap=False.
+            raise RuntimeError(
+                "gradient accumulation detected with hook-overlap enabled; "
+                "construct the optimizer with overlap=False to accumulate "
+                "gradients over multiple backward passes")
         # launch every fully-ready bucket at the head of the fixed schedule
         while self._next < len(self.flat.buckets):
             nb = self.flat.buckets[self._next]

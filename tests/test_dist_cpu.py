@@ -261,3 +261,42 @@ def _async_adam_worker(rank, port, codec, out_file):
 
 def test_async_adam(tmp_path):
     _spawn(_async_adam_worker, None, tmp_path)
+
+
+def _accum_worker(rank, port, codec, out_file):
+    """Gradient accumulation: overlap=False works; overlap=True raises."""
+    from pytorch_ps_mpi_amd import SGD, models
+    _setup(rank, WORLD, port)
+    model, _full, (xs, ys) = _mlp_and_data(rank)
+    opt = SGD(model.named_parameters(), lr=0.05, momentum=0.9,
+              mode="replicated", grad_scale="mean", bucket_mb=0.05,
+              overlap=False)
+    for step in range(3):
+        opt.zero_grad()
+        for _ in range(2):  # 2-step accumulation
+            loss = models.loss_fn("mlp", model, xs, ys)
+            loss.backward()
+        opt.step(loss=loss)
+        assert _checksums_equal(opt)
+    # with overlap=True a second backward must raise
+    model2, _f2, (xs2, ys2) = _mlp_and_data(rank)
+    opt2 = SGD(model2.named_parameters(), lr=0.05, mode="replicated",
+               bucket_mb=0.05, overlap=True)
+    opt2.zero_grad()
+    models.loss_fn("mlp", model2, xs2, ys2).backward()
+    raised = False
+    try:
+        models.loss_fn("mlp", model2, xs2, ys2).backward()
+    except RuntimeError as e:
+        raised = "accumulation" in str(e)
+    assert raised, "expected accumulation guard to fire"
+    # drain the collectives the first backward launched, all ranks together
+    opt2.step()
+    if rank == 0:
+        with open(out_file, "w") as f:
+            f.write("ok")
+    opt.finish()
+
+
+def test_grad_accumulation(tmp_path):
+    _spawn(_accum_worker, None, tmp_path)
