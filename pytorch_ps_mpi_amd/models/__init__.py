@@ -28,7 +28,7 @@ _KIND = {
     "resnet18": ("image", (3, 224, 224), 1000),
     "resnet50": ("image", (3, 224, 224), 1000),
     "vit_b16": ("image", (3, 224, 224), 1000),
-    "gpt2_small": ("tokens", (512,), 50257),
+    "gpt2_small": ("tokens", (512,), 50257),  # targets stay in the real vocab
 }
 
 

@@ -70,5 +70,6 @@ class GPT2(nn.Module):
                                targets.reshape(-1))
 
 
-def gpt2_small(vocab=50257, ctx=1024):
+def gpt2_small(vocab=50304, ctx=1024):
+    # vocab padded 50257 -> 50304 (multiple of 64) for GEMM tile alignment
     return GPT2(vocab=vocab, ctx=ctx)
