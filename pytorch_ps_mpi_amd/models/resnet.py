@@ -9,7 +9,6 @@ channels_last BN kernels were 57% of the step time.
 
 from __future__ import annotations
 
-import torch
 import torch.nn as nn
 
 from ..ops.bn import FusedBatchNorm2d

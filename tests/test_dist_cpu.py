@@ -300,3 +300,39 @@ def _accum_worker(rank, port, codec, out_file):
 
 def test_grad_accumulation(tmp_path):
     _spawn(_accum_worker, None, tmp_path)
+
+
+def _async_skew_worker(rank, port, codec, out_file):
+    """Speed-skewed ranks stress the recv ring / window logic:
+    pass codec == 'slow_ps' (PS sleeps) or 'slow_worker'."""
+    import time
+    from pytorch_ps_mpi_amd import SGD, models
+    _setup(rank, WORLD, port)
+    model, _full, (xs, ys) = _mlp_and_data(rank)
+    opt = SGD(model.named_parameters(), lr=0.02, momentum=0.9, mode="async",
+              bucket_mb=0.05, window=2, max_stale=3)
+    slow_me = (codec == "slow_ps" and rank == 0) or \
+              (codec == "slow_worker" and rank != 0)
+    for step in range(8):
+        if slow_me:
+            time.sleep(0.05)
+        opt.zero_grad()
+        loss = models.loss_fn("mlp", model, xs, ys)
+        loss.backward()
+        l, m = opt.step(loss=loss)
+        assert torch.isfinite(l.detach())
+        if rank != 0:
+            assert m["staleness"] <= 3 + 2, m["staleness"]
+    opt.finish()
+    if rank == 0:
+        assert sum(opt.engine.staleness_hist.values()) >= 6
+        with open(out_file, "w") as f:
+            f.write("ok")
+
+
+def test_async_slow_ps(tmp_path):
+    _spawn(_async_skew_worker, "slow_ps", tmp_path)
+
+
+def test_async_slow_worker(tmp_path):
+    _spawn(_async_skew_worker, "slow_worker", tmp_path)
