@@ -8,6 +8,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops.ce import fused_cross_entropy
 from ..ops.ln import FusedLayerNorm
 
 
@@ -65,8 +66,9 @@ class GPT2(nn.Module):
         # fp32 internally; materializing a fp32 [tokens, vocab] copy cost
         # ~12 ms/step at batch 96 (9.9 GB of extra traffic).
         logits = self.forward(idx)
-        return F.cross_entropy(logits.view(-1, logits.size(-1)),
-                               targets.reshape(-1))
+        return fused_cross_entropy(
+            logits.reshape(-1, logits.size(-1)).contiguous(),
+            targets.reshape(-1))
 
 
 def gpt2_small(vocab=50304, ctx=1024):

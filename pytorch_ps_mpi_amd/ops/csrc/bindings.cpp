@@ -63,6 +63,11 @@ int ps_ln_bwd_dx(void* stream, const void* x, const void* dy, void* dx,
 int ps_ln_bwd_dgb(void* stream, const void* x, const void* dy,
                   const float* mean, const float* rstd, float* dgamma,
                   float* dbeta, int64_t rows, int64_t D);
+int ps_ce_fwd(void* stream, const void* logits, const int64_t* targets,
+              float* losses, float* lse, int64_t T, int64_t V);
+int ps_ce_bwd(void* stream, const void* logits, const int64_t* targets,
+              const float* lse, void* dlogits, int64_t T, int64_t V,
+              float gscale, const float* gout_dev);
 }
 
 namespace {
@@ -397,9 +402,38 @@ void ln_bwd_dgb(at::Tensor x, at::Tensor dy, at::Tensor mean, at::Tensor rstd,
            "ln_bwd_dgb");
 }
 
+void ce_fwd(at::Tensor logits, at::Tensor targets, at::Tensor losses,
+            at::Tensor lse, int64_t T, int64_t V) {
+  TORCH_CHECK(logits.is_cuda() && logits.scalar_type() == at::kBFloat16 &&
+                  logits.is_contiguous(),
+              "logits must be contiguous bf16");
+  TORCH_CHECK(targets.scalar_type() == at::kLong, "targets must be int64");
+  throw_on(ps_ce_fwd(cur_stream(logits), logits.data_ptr(),
+                     targets.data_ptr<int64_t>(), losses.data_ptr<float>(),
+                     lse.data_ptr<float>(), T, V),
+           "ce_fwd");
+}
+
+void ce_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse,
+            at::Tensor dlogits, int64_t T, int64_t V, double gscale,
+            c10::optional<at::Tensor> gout) {
+  const float* gp = nullptr;
+  if (gout.has_value()) {
+    TORCH_CHECK(gout->scalar_type() == at::kFloat && gout->numel() == 1,
+                "gout must be a f32 scalar tensor");
+    gp = gout->data_ptr<float>();
+  }
+  throw_on(ps_ce_bwd(cur_stream(logits), logits.data_ptr(),
+                     targets.data_ptr<int64_t>(), lse.data_ptr<float>(),
+                     dlogits.data_ptr(), T, V, (float)gscale, gp),
+           "ce_bwd");
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("ce_fwd", &ce_fwd);
+  m.def("ce_bwd", &ce_bwd);
   m.def("ln_fwd", &ln_fwd);
   m.def("ln_bwd_dx", &ln_bwd_dx);
   m.def("ln_bwd_dgb", &ln_bwd_dgb);
