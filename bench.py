@@ -108,9 +108,12 @@ def main():
     if on_gpu and world > 1 and dist.is_initialized():
         if rank == 0:
             for _ in range(2):
-                opt.zero_grad()
+                # flat.zero_grad (NOT opt.zero_grad): must not arm the
+                # hook-overlap engine — its collectives would have no
+                # matching calls on the ranks waiting at the barrier
+                opt.flat.zero_grad()
                 models.loss_fn(args.model, model, x, y).backward()
-            opt.zero_grad()
+            opt.flat.zero_grad()
             torch.cuda.synchronize()
         dist.barrier()
 
