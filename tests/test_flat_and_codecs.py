@@ -125,3 +125,43 @@ def test_flatspace_channels_last():
     for (n1, p1), (n2, p2) in zip(m.named_parameters(),
                                   m2.named_parameters()):
         assert torch.allclose(p1.grad, p2.grad, atol=1e-5), n1
+
+
+def test_fused_ce_cpu_fallback():
+    import torch.nn.functional as F
+    from pytorch_ps_mpi_amd.ops.ce import fused_cross_entropy
+    torch.manual_seed(0)
+    logits = torch.randn(16, 64, requires_grad=True)
+    targets = torch.randint(0, 64, (16,))
+    loss = fused_cross_entropy(logits, targets)
+    ref = F.cross_entropy(logits.detach(), targets)
+    assert torch.allclose(loss, ref)
+    loss.backward()
+    assert torch.isfinite(logits.grad).all()
+
+
+def test_fused_ln_cpu_fallback():
+    import torch.nn.functional as F
+    from pytorch_ps_mpi_amd.ops.ln import FusedLayerNorm
+    torch.manual_seed(0)
+    ln = FusedLayerNorm(256)
+    x = torch.randn(4, 256, requires_grad=True)
+    y = ln(x)
+    ref = F.layer_norm(x, (256,), ln.weight, ln.bias, ln.eps)
+    assert torch.allclose(y, ref)
+    y.sum().backward()
+    assert torch.isfinite(x.grad).all()
+
+
+def test_fused_bn_cpu_fallback_module():
+    from pytorch_ps_mpi_amd.ops.bn import FusedBatchNorm2d
+    import torch.nn.functional as F
+    torch.manual_seed(0)
+    bn = FusedBatchNorm2d(8, relu=True)
+    ref = torch.nn.BatchNorm2d(8)
+    with torch.no_grad():
+        ref.weight.copy_(bn.weight)
+        ref.bias.copy_(bn.bias)
+    x = torch.randn(2, 8, 4, 4)
+    y = bn(x)
+    assert torch.allclose(y, F.relu(ref(x)), atol=1e-6)
