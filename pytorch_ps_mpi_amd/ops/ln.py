@@ -50,6 +50,7 @@ class FusedLayerNorm(nn.LayerNorm):
                 and self.normalized_shape[0] % 256 == 0
                 and self.normalized_shape[0] <= 4096
                 and self.elementwise_affine
+                and self.weight is not None and self.bias is not None
                 and self.weight.dtype == torch.bfloat16)
 
     def forward(self, x):

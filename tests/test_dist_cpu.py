@@ -336,3 +336,46 @@ def test_async_slow_ps(tmp_path):
 
 def test_async_slow_worker(tmp_path):
     _spawn(_async_skew_worker, "slow_worker", tmp_path)
+
+
+def _ckpt_worker(rank, port, codec, out_file):
+    import tempfile
+    import torch.distributed as dist
+    from pytorch_ps_mpi_amd import SGD, models
+    from pytorch_ps_mpi_amd.utils import checkpoint
+    _setup(rank, WORLD, port)
+    model, _full, (xs, ys) = _mlp_and_data(rank)
+    opt = SGD(model.named_parameters(), lr=0.05, momentum=0.9,
+              mode="replicated", grad_scale="mean", bucket_mb=0.05)
+    for _ in range(3):
+        opt.zero_grad()
+        loss = models.loss_fn("mlp", model, xs, ys)
+        loss.backward()
+        opt.step(loss=loss)
+    path = os.path.join(tempfile.gettempdir(),
+                        f"ps_ckpt_test_{port}.pt")
+    checkpoint.save(path, opt, extra={"step": 3})  # rank 0 only writes
+    dist.barrier()
+    # fresh model/opt on every rank; rank-0 file restores + re-broadcasts
+    torch.manual_seed(99 + rank)  # deliberately divergent init
+    model2 = models.build_model("mlp")
+    opt2 = SGD(model2.named_parameters(), lr=0.05, momentum=0.9,
+               mode="replicated", grad_scale="mean", bucket_mb=0.05)
+    extra = checkpoint.load(path, opt2)
+    assert extra["step"] == 3
+    assert _checksums_equal(opt2), "ranks inconsistent after restore"
+    assert opt2.flat.param_checksum() == opt.flat.param_checksum()
+    # training continues identically on both optimizers
+    for o, m in ((opt, model), (opt2, model2)):
+        o.zero_grad()
+        models.loss_fn("mlp", m, xs, ys).backward()
+        o.step()
+    assert opt2.flat.param_checksum() == opt.flat.param_checksum()
+    if rank == 0:
+        os.unlink(path)
+        with open(out_file, "w") as f:
+            f.write("ok")
+
+
+def test_checkpoint_distributed(tmp_path):
+    _spawn(_ckpt_worker, None, tmp_path)

@@ -165,3 +165,11 @@ def test_fused_bn_cpu_fallback_module():
     x = torch.randn(2, 8, 4, 4)
     y = bn(x)
     assert torch.allclose(y, F.relu(ref(x)), atol=1e-6)
+
+
+def test_fused_ln_no_bias_fallback():
+    from pytorch_ps_mpi_amd.ops.ln import FusedLayerNorm
+    ln = FusedLayerNorm(256, bias=False)
+    x = torch.randn(4, 256)
+    assert not ln._fast_ok(x)
+    assert torch.isfinite(ln(x)).all()
