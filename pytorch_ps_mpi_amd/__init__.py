@@ -21,8 +21,13 @@ from .codecs import Identity, QuantInt8, TopK, get_codec
 from .optim import PS, SGD, Adam
 from .parallel.comm import Comm, init_distributed
 
-# reference-compatible alias (ps.py:53 class MPI_PS)
-MPI_PS = PS
+def MPI_PS(named_params, *, optim="sgd", code=None, **kwargs):
+    """Reference-compatible constructor (ps.py:53-59): selects the update
+    rule via ``optim='sgd'|'adam'`` and takes the codec as ``code``."""
+    cls = {"sgd": SGD, "adam": Adam}.get(optim)
+    if cls is None:
+        raise ValueError(f"optim must be 'sgd' or 'adam', got {optim!r}")
+    return cls(named_params, code=code, **kwargs)
 
 __all__ = [
     "PS", "MPI_PS", "SGD", "Adam",

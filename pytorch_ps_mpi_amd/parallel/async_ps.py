@@ -327,6 +327,8 @@ class AsyncPSEngine:
         metrics.add("msg_bytes",
                     self.wire_numel * self.wire_dtype.itemsize
                     + flat.total * flat.dtype.itemsize + 24)
+        metrics.add("packaged_bytes",
+                    self.wire_numel * self.wire_dtype.itemsize)
         metrics["staleness"] = self.worker_step - self.last_applied_step
 
     # ---------------------------------------------------------------- step

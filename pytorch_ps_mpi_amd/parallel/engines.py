@@ -153,6 +153,7 @@ class ReplicatedEngine:
         self._works = {}
         self._ready = {}
         metrics.add("msg_bytes", nbytes)
+        metrics.add("packaged_bytes", nbytes)
 
     def finish(self):
         pass
@@ -228,6 +229,7 @@ class SyncPSEngine:
             sum(t.numel() * t.dtype.itemsize
                 for t in self.wire_send.values())
         metrics.add("msg_bytes", wire_b + flat.total * flat.dtype.itemsize)
+        metrics.add("packaged_bytes", wire_b)
 
     def finish(self):
         pass

@@ -152,3 +152,17 @@ def test_print_summary(capsys):
     print_summary(ms)
     outp = capsys.readouterr().out
     assert "optim_step_time" in outp and "mean" in outp
+
+
+def test_mpi_ps_factory():
+    from pytorch_ps_mpi_amd import MPI_PS
+    torch.manual_seed(0)
+    m = models.build_model("mlp")
+    opt = MPI_PS(m.named_parameters(), optim="adam", lr=1e-3)
+    assert isinstance(opt, Adam)
+    m2 = models.build_model("mlp")
+    opt2 = MPI_PS(m2.named_parameters(), optim="sgd", lr=0.1, momentum=0.9)
+    assert isinstance(opt2, SGD)
+    import pytest as _pytest
+    with _pytest.raises(ValueError):
+        MPI_PS(m.named_parameters(), optim="rmsprop")
