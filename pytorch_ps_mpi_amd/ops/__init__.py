@@ -196,3 +196,4 @@ def topk_scatter(dst, idx, val, k, gscale=1.0):
 from . import bn  # noqa: E402,F401  (fused BatchNorm module; needs ops ready)
 from . import ln  # noqa: E402,F401
 from . import ce  # noqa: E402,F401
+from . import attn  # noqa: E402,F401  (experimental)

@@ -68,6 +68,13 @@ int ps_ce_fwd(void* stream, const void* logits, const int64_t* targets,
 int ps_ce_bwd(void* stream, const void* logits, const int64_t* targets,
               const float* lse, void* dlogits, int64_t T, int64_t V,
               float gscale, const float* gout_dev);
+int ps_attn_fwd(void* stream, const void* q, const void* k, const void* v,
+                void* o, float* lse, int64_t rows, int64_t N, float scale,
+                int causal);
+int ps_attn_bwd(void* stream, const void* q, const void* k, const void* v,
+                const void* o, const void* dout, const float* lse,
+                float* delta, void* dq, void* dk, void* dv, int64_t rows,
+                int64_t N, float scale, int causal);
 }
 
 namespace {
@@ -431,7 +438,34 @@ void ce_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse,
 
 }  // namespace
 
+void attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o,
+              at::Tensor lse, int64_t N, double scale, bool causal) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16 &&
+                  q.is_contiguous(),
+              "q must be contiguous bf16");
+  const int64_t rows = q.numel() / 64;
+  throw_on(ps_attn_fwd(cur_stream(q), q.data_ptr(), k.data_ptr(),
+                       v.data_ptr(), o.data_ptr(), lse.data_ptr<float>(),
+                       rows, N, (float)scale, causal ? 1 : 0),
+           "attn_fwd");
+}
+
+void attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o,
+              at::Tensor dout, at::Tensor lse, at::Tensor delta,
+              at::Tensor dq, at::Tensor dk, at::Tensor dv, int64_t N,
+              double scale, bool causal) {
+  const int64_t rows = q.numel() / 64;
+  throw_on(ps_attn_bwd(cur_stream(q), q.data_ptr(), k.data_ptr(),
+                       v.data_ptr(), o.data_ptr(), dout.data_ptr(),
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),
+                       dq.data_ptr(), dk.data_ptr(), dv.data_ptr(), rows, N,
+                       (float)scale, causal ? 1 : 0),
+           "attn_bwd");
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("attn_fwd", &attn_fwd);
+  m.def("attn_bwd", &attn_bwd);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);
   m.def("ln_fwd", &ln_fwd);
