@@ -1,9 +1,9 @@
-"""EXPERIMENTAL fused attention (ops/csrc/attn_kernels.hip).
+"""Fused flash attention (ops/csrc/attn_kernels.hip).
 
-Round-2 base: correctness-first flash fwd+bwd, one wave per row, D == 64.
-NOT used by the models yet (they call torch SDPA); the GPU numerics test is
-gated behind PS_EXPERIMENTAL=1 until hardware-validated.  ROADMAP.md item 2
-is the MFMA-tiled rewrite.
+Correctness-first flash fwd+bwd, one wave per row, D == 64 — hardware-
+validated against torch SDPA (tests/test_gpu_attn.py) but not yet
+perf-competitive with aotriton, so the models keep torch SDPA.  ROADMAP.md
+item 2 is the MFMA-tiled rewrite of the inner loops.
 """
 
 from __future__ import annotations

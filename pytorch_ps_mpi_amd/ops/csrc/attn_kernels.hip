@@ -1,9 +1,8 @@
-// attn_kernels.hip — EXPERIMENTAL flash-attention fwd+bwd for CDNA4.
+// attn_kernels.hip — flash-attention fwd+bwd for CDNA4 (gfx950).
 //
-// Status: correctness-first round-2 base (compile-checked; the GPU numerics
-// test is env-gated behind PS_EXPERIMENTAL until validated on hardware).
-// NOT wired into the models — they use torch SDPA until this is validated
-// and MFMA-tiled (see ROADMAP.md item 2).
+// Status: correctness-first; hardware-validated against torch SDPA
+// (tests/test_gpu_attn.py) but not yet perf-competitive with aotriton —
+// the models keep torch SDPA until the MFMA-tiled rewrite (ROADMAP item 2).
 //
 // Shape contract: q,k,v,o,do,dq,dk,dv are [B, H, N, D] bf16 contiguous with
 // D == 64 (one lane per head dim).  Online-softmax forward saves
