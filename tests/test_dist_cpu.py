@@ -379,3 +379,39 @@ def _ckpt_worker(rank, port, codec, out_file):
 
 def test_checkpoint_distributed(tmp_path):
     _spawn(_ckpt_worker, None, tmp_path)
+
+
+class _PartialUseModel(torch.nn.Module):
+    """A parameter whose hook never fires (unused in forward)."""
+
+    def __init__(self):
+        super().__init__()
+        self.used = torch.nn.Linear(8, 4)
+        self.unused = torch.nn.Linear(8, 4)
+
+    def forward(self, x):
+        return self.used(x)
+
+
+def _unused_param_worker(rank, port, codec, out_file):
+    import torch.nn.functional as F
+    from pytorch_ps_mpi_amd import SGD
+    _setup(rank, WORLD, port)
+    torch.manual_seed(0)
+    model = _PartialUseModel()
+    opt = SGD(model.named_parameters(), lr=0.05, momentum=0.9,
+              mode="replicated", grad_scale="mean", bucket_mb=0.05)
+    x = torch.randn(8, 8)
+    y = torch.randint(0, 4, (8,))
+    for _ in range(3):
+        opt.zero_grad()
+        F.cross_entropy(model(x), y).backward()
+        opt.step()  # unused param's bucket launches at step() time
+        assert _checksums_equal(opt)
+    if rank == 0:
+        with open(out_file, "w") as f:
+            f.write("ok")
+
+
+def test_unused_param_bucket(tmp_path):
+    _spawn(_unused_param_worker, None, tmp_path)

@@ -173,3 +173,39 @@ def test_fused_ln_no_bias_fallback():
     x = torch.randn(4, 256)
     assert not ln._fast_ok(x)
     assert torch.isfinite(ln(x)).all()
+
+
+def test_topk_density_one_and_tiny():
+    c = codecs.TopK(density=1.0)
+    n = 40
+    src = torch.randn(n)
+    wire = torch.zeros(c.wire_numel(n, torch.float32), dtype=torch.uint8)
+    c.encode(src, wire)
+    dst = torch.zeros(n)
+    c.decode_reduce(dst, [wire], src_dtype=torch.float32)
+    assert torch.allclose(dst, src, atol=1e-6)
+    # min_k clamp: k never exceeds numel
+    assert codecs.TopK(density=0.001, min_k=100).k_for(7) == 7
+
+
+def test_quant8_tiny_chunk():
+    c = codecs.QuantInt8()
+    n = 13  # < one chunk
+    src = torch.randn(n)
+    wire = torch.zeros(c.wire_numel(n), dtype=torch.uint8)
+    c.encode(src, wire)
+    dst = torch.zeros(n)
+    c.decode_reduce(dst, [wire])
+    assert (dst - src).abs().max() < 0.05
+
+
+def test_flatspace_skips_frozen_params():
+    import torch.nn as nn
+    m = nn.Sequential(nn.Linear(8, 8), nn.Linear(8, 4))
+    m[0].weight.requires_grad_(False)
+    flat = FlatSpace(m.named_parameters(), bucket_elems=1 << 20)
+    names = [e[0] for e in flat.entries]
+    assert "0.weight" not in names and "1.weight" in names
+    y = m(torch.randn(2, 8)).sum()
+    y.backward()
+    assert flat.flat_grad.abs().sum() > 0
