@@ -154,6 +154,93 @@ class QuantInt8:
         return self.wire_numel(numel)
 
 
+class HostCodec:
+    """Adapter for reference-style codec OBJECTS (ps.py:18).
+
+    Wraps a `codings`-style plugin with the reference contract
+        code.encode(grad_array, **kw) -> picklable object
+        code.decode(obj)              -> array-like gradient
+        code.codes = [...]            (stashed before decoding, ps.py:165)
+    into this framework's fixed-capacity wire contract.  The payload is
+    pickled host bytes with a 4-byte length header — i.e. the reference's
+    own wire format (mpi_comms.py:186-193) made RCCL-safe: capacity is fixed
+    per bucket (sized from a deterministic dry-run encode of a zero gradient
+    x `headroom`, floor `min_capacity` — the reference used 10x and 15 KiB,
+    mpi_comms.py:82-83) and overflow raises loudly instead of a sentinel
+    scan.  Host round trips make this a compatibility path, not a fast one:
+    use the device codecs (TopK/QuantInt8) for production.
+    """
+
+    name = "host"
+    supports_allreduce = False
+
+    def __init__(self, code, headroom=10.0, min_capacity=15 * 1024):
+        self.code = code
+        self.headroom = headroom
+        self.min_capacity = min_capacity
+        self._cap = {}
+
+    def wire_dtype(self, src_dtype):
+        return torch.uint8
+
+    def _capacity(self, numel):
+        if numel not in self._cap:
+            import pickle
+            probe = self.code.encode(
+                torch.zeros(numel, dtype=torch.float32).numpy())
+            need = len(pickle.dumps(probe, protocol=4)) + 4
+            cap = max(self.min_capacity, int(need * self.headroom))
+            self._cap[numel] = (cap + 15) // 16 * 16
+        return self._cap[numel]
+
+    def wire_numel(self, numel, src_dtype=None):
+        return self._capacity(numel)
+
+    def encode(self, src, wire):
+        import pickle
+        obj = self.code.encode(src.detach().float().cpu().numpy())
+        blob = pickle.dumps(obj, protocol=4)
+        if len(blob) + 4 > wire.numel():
+            raise RuntimeError(
+                f"HostCodec payload {len(blob)}B exceeds wire capacity "
+                f"{wire.numel()}B — raise headroom= (reference overflowed "
+                "its sentinel here, mpi_comms.py:96-104)")
+        import numpy as np
+        wire[:4].copy_(torch.from_numpy(
+            np.frombuffer(np.int32(len(blob)).tobytes(), dtype=np.uint8)
+            .copy()))
+        wire[4:4 + len(blob)].copy_(torch.from_numpy(
+            np.frombuffer(blob, dtype=np.uint8).copy()).to(wire.device))
+
+    def decode_reduce(self, dst, wires, gscale=1.0, beta=0.0, src_dtype=None):
+        import pickle
+
+        import numpy as np
+        objs = []
+        for w in wires:
+            wc = w.cpu().numpy()
+            n = int(np.frombuffer(wc[:4].tobytes(), dtype=np.int32)[0])
+            objs.append(pickle.loads(wc[4:4 + n].tobytes()))
+        # reference contract: stash raw codes before decoding (ps.py:165)
+        try:
+            self.code.codes = objs
+        except Exception:
+            pass
+        acc = None
+        for obj in objs:
+            g = self.code.decode(obj)
+            t = torch.as_tensor(np.asarray(g), dtype=torch.float32).reshape(-1)
+            acc = t if acc is None else acc + t
+        acc = acc.to(dst.device)
+        if beta == 0.0:
+            dst.copy_(acc * gscale)
+        else:
+            dst.mul_(beta).add_(acc, alpha=gscale)
+
+    def bytes_on_wire(self, numel, dtype=None):
+        return self._capacity(numel)
+
+
 def get_codec(spec):
     """'identity' | 'topk' | 'topk:0.02' | 'quant8' | codec instance | None."""
     if spec is None:

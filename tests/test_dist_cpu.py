@@ -415,3 +415,59 @@ def _unused_param_worker(rank, port, codec, out_file):
 
 def test_unused_param_bucket(tmp_path):
     _spawn(_unused_param_worker, None, tmp_path)
+
+
+class _RefStyleTopK:
+    """A reference-style `codings` plugin: encode -> picklable dict,
+    decode -> ndarray (SURVEY §2.2 contract)."""
+
+    def __init__(self, k=64):
+        self.k = k
+        self.codes = None
+
+    def encode(self, grad, **kw):
+        import numpy as np
+        flat = np.asarray(grad).reshape(-1)
+        idx = np.argsort(np.abs(flat))[-self.k:]
+        return {"n": flat.size, "idx": idx.astype("int32"),
+                "val": flat[idx].astype("float32")}
+
+    def decode(self, obj, **kw):
+        import numpy as np
+        out = np.zeros(obj["n"], dtype="float32")
+        out[obj["idx"]] = obj["val"]
+        return out
+
+
+def _host_codec_worker(rank, port, codec, out_file):
+    from pytorch_ps_mpi_amd import SGD, HostCodec, models
+    _setup(rank, WORLD, port)
+    model, _full, (xs, ys) = _mlp_and_data(rank)
+    user_code = _RefStyleTopK(k=512)
+    opt = SGD(model.named_parameters(), lr=0.05, momentum=0.9,
+              mode=codec,  # "replicated" or "ps"
+              code=HostCodec(user_code), grad_scale="mean", bucket_mb=0.05)
+    losses = []
+    for _ in range(5):
+        opt.zero_grad()
+        loss = models.loss_fn("mlp", model, xs, ys)
+        loss.backward()
+        l, m = opt.step(loss=loss)
+        losses.append(float(l.detach()))
+        assert _checksums_equal(opt)
+    assert losses[-1] < losses[0]
+    # the reference stashes raw codes on the object before decoding
+    if codec == "replicated" or rank == 0:
+        assert user_code.codes is not None and len(user_code.codes) == WORLD
+    if rank == 0:
+        with open(out_file, "w") as f:
+            f.write("ok")
+    opt.finish()
+
+
+def test_host_codec_replicated(tmp_path):
+    _spawn(_host_codec_worker, "replicated", tmp_path)
+
+
+def test_host_codec_sync_ps(tmp_path):
+    _spawn(_host_codec_worker, "ps", tmp_path)
