@@ -471,3 +471,28 @@ def test_host_codec_replicated(tmp_path):
 
 def test_host_codec_sync_ps(tmp_path):
     _spawn(_host_codec_worker, "ps", tmp_path)
+
+
+def _host_codec_async_worker(rank, port, codec, out_file):
+    from pytorch_ps_mpi_amd import SGD, HostCodec
+    from pytorch_ps_mpi_amd import models
+    _setup(rank, WORLD, port)
+    model, _full, (xs, ys) = _mlp_and_data(rank)
+    opt = SGD(model.named_parameters(), lr=0.02, momentum=0.9, mode="async",
+              code=HostCodec(_RefStyleTopK(k=512)), bucket_mb=0.05,
+              window=2, max_stale=4)
+    for _ in range(6):
+        opt.zero_grad()
+        loss = models.loss_fn("mlp", model, xs, ys)
+        loss.backward()
+        l, _ = opt.step(loss=loss)
+        assert torch.isfinite(l.detach())
+    opt.finish()
+    if rank == 0:
+        assert sum(opt.engine.staleness_hist.values()) >= 4
+        with open(out_file, "w") as f:
+            f.write("ok")
+
+
+def test_host_codec_async(tmp_path):
+    _spawn(_host_codec_async_worker, None, tmp_path)

@@ -110,3 +110,21 @@ def test_checkpoint_gpu(tmp_path):
     checkpoint.load(str(tmp_path / "ck.pt"), opt2)
     assert torch.equal(opt.flat.flat_param, opt2.flat.flat_param)
     assert torch.equal(opt._mom, opt2._mom)
+
+
+def test_profile_gpu_metrics():
+    """profile_gpu=True adds HIP-event *_gpu_ms spans to the metrics dict."""
+    device = torch.device("cuda:0")
+    torch.manual_seed(0)
+    model = models.build_model("mlp", device=device, dtype=torch.bfloat16)
+    opt = SGD(model.named_parameters(), lr=0.1, momentum=0.9,
+              profile_gpu=True)
+    x, y = models.synthetic_batch("mlp", 8, device=device,
+                                  dtype=torch.bfloat16, seed=0)
+    opt.zero_grad()
+    loss = models.loss_fn("mlp", model, x, y)
+    loss.backward()
+    _, m = opt.step(loss=loss)
+    gpu_keys = [k for k in m if k.endswith("_gpu_ms")]
+    assert gpu_keys, m.keys()
+    assert all(v >= 0.0 for k, v in m.items() if k.endswith("_gpu_ms"))
