@@ -37,7 +37,8 @@ class PS(torch.optim.Optimizer):
     def __init__(self, named_params, defaults, *, code=None, mode="replicated",
                  bucket_mb=50, grad_scale="sum", window=2, max_stale=8,
                  quorum=1, dedicated_ps=False, dtype=None, overlap=True,
-                 debug_consistency=0, profile_gpu=False):
+                 debug_consistency=0, profile_gpu=False, reply_shards="auto",
+                 serve_timeout_s=None):
         named_params = list(named_params)
         if named_params and not isinstance(named_params[0], tuple):
             raise TypeError("pass model.named_parameters(), not parameters()")
@@ -67,7 +68,9 @@ class PS(torch.optim.Optimizer):
             self.engine = AsyncPSEngine(self.flat, self.codec, self.comm,
                                         grad_scale=gscale, window=window,
                                         max_stale=max_stale, quorum=quorum,
-                                        dedicated=dedicated_ps)
+                                        dedicated=dedicated_ps,
+                                        reply_shards=reply_shards,
+                                        serve_timeout_s=serve_timeout_s)
         elif self.comm.world <= 1:
             self.engine = LocalEngine(self.flat, self.codec, self.comm, gscale)
         elif mode == "replicated":
@@ -155,9 +158,15 @@ class PS(torch.optim.Optimizer):
                 getattr(self.engine, "staleness_hist", {}))
         return loss, dict(metrics)
 
-    def finish(self):
-        """Drain async traffic; call once after the training loop."""
-        self.engine.finish()
+    def finish(self, barrier=True):
+        """Drain async traffic; call once after the training loop.
+
+        ``barrier=False`` skips the final cross-rank barrier — required when
+        a peer may have been dropped (it can never reach the barrier)."""
+        if isinstance(self.engine, AsyncPSEngine):
+            self.engine.finish(barrier=barrier)
+        else:
+            self.engine.finish()
 
     def serve(self):
         """Dedicated-PS event loop (mode='async', dedicated_ps=True)."""
