@@ -164,20 +164,29 @@ class HostCodec:
     into this framework's fixed-capacity wire contract.  The payload is
     pickled host bytes with a 4-byte length header — i.e. the reference's
     own wire format (mpi_comms.py:186-193) made RCCL-safe: capacity is fixed
-    per bucket (sized from a deterministic dry-run encode of a zero gradient
-    x `headroom`, floor `min_capacity` — the reference used 10x and 15 KiB,
-    mpi_comms.py:82-83) and overflow raises loudly instead of a sentinel
-    scan.  Host round trips make this a compatibility path, not a fast one:
-    use the device codecs (TopK/QuantInt8) for production.
+    per bucket, sized from a deterministic dry-run encode of a REPRESENTATIVE
+    seeded-random gradient × `headroom` (floor `min_capacity` — the reference
+    used 10× headroom and a 15 KiB floor over a running max of real sizes,
+    mpi_comms.py:82-83).  The random probe, not a zero probe, so adaptive /
+    entropy-coding plugins whose payload grows with content are sized
+    realistically (advisor round-1 finding); all ranks compute the same probe
+    (fixed seed), so capacities agree without a size exchange.  OVERFLOW
+    ABORTS THE RUN with a loud error (no mid-training capacity renegotiation:
+    recv slots are pre-posted at fixed sizes) — raise `headroom=` or pass an
+    explicit `capacity=` for plugins with unbounded worst cases.  Host round
+    trips make this a compatibility path, not a fast one: use the device
+    codecs (TopK/QuantInt8) for production.
     """
 
     name = "host"
     supports_allreduce = False
 
-    def __init__(self, code, headroom=10.0, min_capacity=15 * 1024):
+    def __init__(self, code, headroom=10.0, min_capacity=15 * 1024,
+                 capacity=None):
         self.code = code
         self.headroom = headroom
         self.min_capacity = min_capacity
+        self.capacity = capacity  # explicit per-bucket byte override
         self._cap = {}
 
     def wire_dtype(self, src_dtype):
@@ -185,9 +194,13 @@ class HostCodec:
 
     def _capacity(self, numel):
         if numel not in self._cap:
+            if self.capacity is not None:
+                self._cap[numel] = (int(self.capacity) + 15) // 16 * 16
+                return self._cap[numel]
             import pickle
-            probe = self.code.encode(
-                torch.zeros(numel, dtype=torch.float32).numpy())
+            g = torch.Generator().manual_seed(0x5eed ^ numel)
+            probe_grad = torch.randn(numel, generator=g) * 1e-2
+            probe = self.code.encode(probe_grad.numpy())
             need = len(pickle.dumps(probe, protocol=4)) + 4
             cap = max(self.min_capacity, int(need * self.headroom))
             self._cap[numel] = (cap + 15) // 16 * 16

@@ -90,7 +90,11 @@ k_fused_sgd(float* __restrict__ p, float* __restrict__ buf,
 }
 
 // ---------------------------------------------------------------------------
-// fused Adam / AdamW-free (reference Adam, ps.py:218-261, = torch.optim.Adam)
+// fused Adam incl. amsgrad — matches MODERN torch.optim.Adam exactly:
+//   denom = sqrt(v)/sqrt(bc2) + eps.
+// The reference (ps.py:218-261, torch-0.3-era) used sqrt(v) + eps with the
+// bias correction folded into step_size, which effectively scales eps by
+// sqrt(bc2): a benign O(eps) deviation from ps.py, not bit-identical to it.
 // ---------------------------------------------------------------------------
 
 template <bool AMSGRAD, typename OUT_T>

@@ -244,7 +244,13 @@ class SGD(PS):
 
 
 class Adam(PS):
-    """Reference Adam.optim_step semantics (ps.py:217-261), fused on-device."""
+    """Adam incl. amsgrad (reference Adam.optim_step, ps.py:217-261), fused
+    on-device.  Eps placement follows MODERN torch.optim.Adam:
+    ``denom = sqrt(v)/sqrt(bias_correction2) + eps``; the reference (like the
+    torch of its era) used ``sqrt(v) + eps`` with the correction folded into
+    step_size, which scales eps by sqrt(bc2) — a benign O(eps) deviation, but
+    not bit-identical to ps.py:257-261.  Matches torch.optim.Adam exactly
+    (parity test: tests/test_optim_local.py)."""
 
     def __init__(self, named_params, lr=1e-3, betas=(0.9, 0.999), eps=1e-8,
                  weight_decay=0.0, amsgrad=False, **ps_kwargs):

@@ -337,18 +337,27 @@ class AsyncPSEngine:
         self._post(st, slot)
         st.head = (st.head + 1) % self.ring
 
-    def poll_serve(self, metrics, max_per_peer=None):
-        """Serve arrived pushes without blocking. PS-side only."""
+    def poll_serve(self, metrics, max_per_peer=None, drop_on_error=False):
+        """Serve arrived pushes without blocking. PS-side only.
+
+        With drop_on_error (dedicated serve loop) an exception while serving
+        one peer — e.g. gloo "connection closed by peer" from a dead worker —
+        retires THAT peer and serving continues for the rest."""
         self._start_ps()
         served = 0
         budget = max_per_peer if max_per_peer is not None else self.ring
         for st in self.peers.values():
             n = 0
-            while (not st.stopped and n < budget
-                   and _tagged(st.slots[st.head])):
-                self._serve_slot(st, metrics)
-                served += 1
-                n += 1
+            try:
+                while (not st.stopped and n < budget
+                       and _tagged(st.slots[st.head])):
+                    self._serve_slot(st, metrics)
+                    served += 1
+                    n += 1
+            except Exception as e:
+                if not drop_on_error:
+                    raise
+                self._drop_peer(st, f"serve error: {e!r}")
         return served
 
     def serve(self, metrics):
@@ -361,14 +370,7 @@ class AsyncPSEngine:
             alive = [st for st in self.peers.values() if not st.stopped]
             if not alive:
                 break
-            try:
-                progressed = self.poll_serve(metrics)
-            except Exception as e:  # a dying peer can surface here (gloo
-                # connection reset) — retire the noisiest candidate
-                victim = max(alive, key=lambda s: time.monotonic()
-                             - s.last_seen)
-                self._drop_peer(victim, f"serve error: {e!r}")
-                continue
+            progressed = self.poll_serve(metrics, drop_on_error=True)
             if progressed == 0:
                 now = time.monotonic()
                 if self.serve_timeout_s is not None:

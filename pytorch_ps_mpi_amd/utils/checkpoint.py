@@ -28,6 +28,10 @@ def load(path, optimizer):
     payload = torch.load(path, map_location="cpu", weights_only=False)
     optimizer.load_state_dict(payload["optimizer"])
     if optimizer.comm.initialized:
-        dist.broadcast(optimizer.flat.flat_param, src=0)
-        optimizer.flat.sync_master_from_param()
+        # Broadcast the fp32 MASTER (not the bf16 model copy) and re-derive
+        # flat_param from it: broadcasting bf16 then syncing master FROM it
+        # would round the master through bf16 and a resumed run would diverge
+        # from an uninterrupted one (advisor round-1 finding).
+        dist.broadcast(optimizer.flat.master, src=0)
+        optimizer.flat.sync_param_from_master()
     return payload.get("extra", {})

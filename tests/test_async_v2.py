@@ -1,0 +1,178 @@
+"""Round-2 async-PS protocol tests (gloo, CPU):
+
+  * bucket-pipelined pushes really launch from backward hooks (the push is
+    in flight before step() is called),
+  * sharded replies cover the whole parameter space round-robin,
+  * a dedicated PS survives a worker that dies silently (peer drop),
+  * staleness accounting is measured from push time.
+"""
+
+import os
+import time
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+pytestmark = pytest.mark.timeout(600)
+
+
+def _free_port():
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _setup(rank, world, port):
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["LOCAL_RANK"] = str(rank)
+    from pytorch_ps_mpi_amd import init_distributed
+    return init_distributed(backend="gloo")
+
+
+# ---------------------------------------------------------------- pipelining
+
+def _pipelined_worker(rank, port, out_file):
+    import torch.nn as nn
+    import torch.nn.functional as F
+    from pytorch_ps_mpi_amd import SGD
+    _setup(rank, 2, port)
+    torch.manual_seed(0)
+    # deep narrow stack -> many similar-size buckets at tiny bucket_mb
+    model = nn.Sequential(*[nn.Sequential(nn.Linear(64, 64), nn.ReLU())
+                            for _ in range(6)], nn.Linear(64, 10))
+    g = torch.Generator().manual_seed(rank + 1)
+    x = torch.randn(16, 64, generator=g)
+    y = torch.randint(0, 10, (16,), generator=g)
+    class models:  # same call shape as pytorch_ps_mpi_amd.models.loss_fn
+        @staticmethod
+        def loss_fn(_n, m, xx, yy):
+            return F.cross_entropy(m(xx), yy)
+    # tiny buckets -> many buckets -> several reply shards
+    opt = SGD(model.named_parameters(), lr=0.02, momentum=0.9, mode="async",
+              bucket_mb=0.01, window=2, max_stale=4)
+    eng = opt.engine
+    if rank != 0:
+        assert len(opt.flat.buckets) >= 4, "test needs multiple buckets"
+        assert eng.n_shards == 2
+        # shards tile the flat space exactly
+        assert eng.shards[0][0] == 0
+        assert eng.shards[-1][1] == opt.flat.total
+        for (a, b), (c, d) in zip(eng.shards, eng.shards[1:]):
+            assert b == c
+    losses = []
+    for step in range(10):
+        opt.zero_grad()
+        loss = models.loss_fn("mlp", model, x, y)
+        loss.backward()
+        if rank != 0:
+            # the push must already be in flight from the hooks: header sent
+            # and at least one bucket launched DURING backward
+            assert eng._cur is not None, "hooks did not start the push"
+            assert eng._next >= 1, "no bucket launched during backward"
+        l, m = opt.step(loss=loss)
+        losses.append(float(l.detach()))
+        if rank != 0:
+            assert m["staleness"] <= 4 + 2
+    opt.finish()
+    assert all(torch.isfinite(torch.tensor(losses)))
+    assert losses[-1] < losses[0]
+    if rank != 0:
+        # every shard was refreshed at least once (full round-robin coverage)
+        assert eng.cursor >= eng.n_shards
+        with open(out_file, "w") as f:
+            f.write("ok")
+
+
+def test_async_pipelined_multibucket(tmp_path):
+    out = str(tmp_path / "ok.txt")
+    mp.spawn(_pipelined_worker, args=(_free_port(), out), nprocs=2, join=True)
+    assert os.path.exists(out)
+
+
+# ---------------------------------------------------------------- peer drop
+
+def _drop_worker(rank, port, out_file):
+    from pytorch_ps_mpi_amd import SGD, models
+    _setup(rank, 3, port)
+    torch.manual_seed(0)
+    model = models.build_model("mlp")
+    opt = SGD(model.named_parameters(), lr=0.02, momentum=0.9, mode="async",
+              bucket_mb=0.05, window=2, max_stale=4, dedicated_ps=True,
+              serve_timeout_s=2.0)
+    if rank == 0:
+        opt.serve()
+        eng = opt.engine
+        assert eng.peers_dropped == 1, eng.peers_dropped
+        # worker 1's 8 pushes were all served despite worker 2 dying
+        assert sum(eng.staleness_hist.values()) >= 8 + 2
+        opt.finish(barrier=False)
+        with open(out_file, "w") as f:
+            f.write("ok")
+        os._exit(0)  # pending recvs to the dead peer: skip pg teardown
+    elif rank == 1:
+        x, y = models.synthetic_batch("mlp", 8, seed=rank)
+        for _ in range(8):
+            opt.zero_grad()
+            loss = models.loss_fn("mlp", model, x, y)
+            loss.backward()
+            opt.step(loss=loss)
+        opt.finish(barrier=False)
+        os._exit(0)
+    else:
+        # rank 2: pushes twice, then dies without stop markers
+        x, y = models.synthetic_batch("mlp", 8, seed=rank)
+        for _ in range(2):
+            opt.zero_grad()
+            loss = models.loss_fn("mlp", model, x, y)
+            loss.backward()
+            opt.step(loss=loss)
+        time.sleep(0.2)  # let the sends drain
+        os._exit(0)
+
+
+def test_async_peer_drop(tmp_path):
+    out = str(tmp_path / "ok.txt")
+    mp.spawn(_drop_worker, args=(_free_port(), out), nprocs=3, join=True)
+    assert os.path.exists(out)
+
+
+# ------------------------------------------------- staleness from push time
+
+def _stale_worker(rank, port, out_file):
+    from pytorch_ps_mpi_amd import SGD, models
+    _setup(rank, 2, port)
+    torch.manual_seed(0)
+    model = models.build_model("mlp")
+    x, y = models.synthetic_batch("mlp", 8, seed=rank + 1)
+    opt = SGD(model.named_parameters(), lr=0.02, momentum=0.9, mode="async",
+              bucket_mb=0.05, window=2, max_stale=3)
+    eng = opt.engine
+    for step in range(8):
+        opt.zero_grad()
+        loss = models.loss_fn("mlp", model, x, y)
+        loss.backward()
+        l, m = opt.step(loss=loss)
+        if rank != 0:
+            # last_applied_step is a PUSH step that was actually sent, and
+            # staleness = worker_step - that push's step
+            assert eng.last_applied_step <= eng.worker_step
+            assert m["staleness"] == eng.worker_step - eng.last_applied_step
+            assert m["staleness"] <= 3 + 2
+    opt.finish()
+    if rank != 0:
+        assert eng.last_applied_step > 0, "no reply was ever applied"
+        with open(out_file, "w") as f:
+            f.write("ok")
+
+
+def test_async_staleness_from_push_time(tmp_path):
+    out = str(tmp_path / "ok.txt")
+    mp.spawn(_stale_worker, args=(_free_port(), out), nprocs=2, join=True)
+    assert os.path.exists(out)

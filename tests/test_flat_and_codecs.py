@@ -209,3 +209,50 @@ def test_flatspace_skips_frozen_params():
     y = m(torch.randn(2, 8)).sum()
     y.backward()
     assert flat.flat_grad.abs().sum() > 0
+
+
+def test_host_codec_adaptive_capacity():
+    """A content-adaptive plugin (payload ~ #nonzeros) must not overflow the
+    fixed wire capacity: HostCodec sizes capacity from a seeded RANDOM probe
+    (a zero-gradient probe would have sized it at the 15 KiB floor and a real
+    dense gradient would abort mid-training — advisor round-1 finding)."""
+    import numpy as np
+
+    class SparseAdaptive:
+        """Stores (idx, val) pairs of nonzero entries — size tracks content."""
+
+        def encode(self, arr):
+            nz = np.flatnonzero(arr)
+            return (nz.astype(np.int32), arr[nz].astype(np.float32))
+
+        def decode(self, obj):
+            idx, val = obj
+            n = 20000
+            out = np.zeros(n, dtype=np.float32)
+            out[idx] = val
+            return out
+
+    c = codecs.HostCodec(SparseAdaptive(), headroom=1.5)
+    n = 20000
+    cap = c.wire_numel(n)
+    # capacity reflects a dense payload (8B/elem + pickle framing), not the
+    # 15 KiB zero-probe floor
+    assert cap > 8 * n
+    src = torch.randn(n)  # fully dense — worst case for this plugin
+    wire = torch.zeros(cap, dtype=torch.uint8)
+    c.encode(src, wire)
+    dst = torch.zeros(n)
+    c.decode_reduce(dst, [wire], src_dtype=torch.float32)
+    assert torch.allclose(src, dst, atol=1e-6)
+
+
+def test_host_codec_explicit_capacity():
+    class Dense:
+        def encode(self, arr):
+            return arr
+
+        def decode(self, obj):
+            return obj
+
+    c = codecs.HostCodec(Dense(), capacity=1 << 20)
+    assert c.wire_numel(123) == 1 << 20
