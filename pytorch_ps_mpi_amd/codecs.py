@@ -117,7 +117,10 @@ class TopK:
 class QuantInt8:
     """Per-256-element-chunk absmax int8 quantization.
 
-    Wire layout (uint8): [ f32 scales[nchunks] | int8 q[numel] | pad ].
+    Wire layout (uint8): [ f32 scales[nchunks] | pad to 16B | int8 q[numel]
+    | pad ] — the pad keeps the int8 payload 16B-aligned so the HIP kernels
+    move it with 8B packed loads/stores (scalar int8 I/O was the round-1
+    bandwidth ceiling).
     """
 
     name = "quant8"
@@ -126,15 +129,20 @@ class QuantInt8:
     def wire_dtype(self, src_dtype):
         return torch.uint8
 
+    @staticmethod
+    def _qoff(nc):
+        return (4 * nc + 15) // 16 * 16
+
     def wire_numel(self, numel, src_dtype=None):
         nc = ops.quant8_nscales(numel)
-        raw = 4 * nc + numel
+        raw = self._qoff(nc) + numel
         return (raw + 15) // 16 * 16
 
     def _views(self, wire, numel):
         nc = ops.quant8_nscales(numel)
+        qoff = self._qoff(nc)
         scales = wire[:4 * nc].view(torch.float32)
-        q = wire[4 * nc:4 * nc + numel].view(torch.int8)
+        q = wire[qoff:qoff + numel].view(torch.int8)
         return scales, q
 
     def encode(self, src, wire):

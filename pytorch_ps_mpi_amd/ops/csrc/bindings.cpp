@@ -92,6 +92,10 @@ void check_flat(const at::Tensor& t, at::ScalarType dt, const char* name) {
 int is_bf16(const at::Tensor& t) { return t.scalar_type() == at::kBFloat16 ? 1 : 0; }
 
 void throw_on(int err, const char* what) {
+  TORCH_CHECK(err != 9001, what, ": 1..8 sources supported");
+  TORCH_CHECK(err != 9002, what,
+              ": int8 payload/dst must be 8/16B-aligned (use the codec wire "
+              "layout: scales padded to 16B before q)");
   TORCH_CHECK(err == 0, what, ": hip error ", err);
 }
 
