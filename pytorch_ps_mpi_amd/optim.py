@@ -82,9 +82,14 @@ class PS(torch.optim.Optimizer):
         else:
             raise ValueError(f"unknown mode {mode!r}")
 
-        # all ranks start from rank 0's parameters
+        # all ranks start from rank 0's parameters.  The async engine uses
+        # per-peer p2p (collective-free control plane); the collective
+        # engines use a broadcast.
         if self.comm.initialized:
-            dist.broadcast(self.flat.flat_param, src=0)
+            if isinstance(self.engine, AsyncPSEngine):
+                self.engine.initial_param_sync()
+            else:
+                dist.broadcast(self.flat.flat_param, src=0)
             self.flat.sync_master_from_param()
 
         # backward-hook comm overlap (replaces the reference's 200-thread

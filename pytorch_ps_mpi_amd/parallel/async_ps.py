@@ -254,6 +254,23 @@ class AsyncPSEngine:
                 lo = b.end
         return shards
 
+    def initial_param_sync(self):
+        """Deliver rank 0's initial parameters over the per-peer reply
+        channels (p2p) instead of a default-group broadcast.  This keeps the
+        whole async engine collective-free, which (a) matches the PS pattern
+        — xGMI is point-to-point, rank 0 reaches each peer on its own link —
+        and (b) lets world>1 RCCL tests run with two ranks sharing one GPU
+        (RCCL rejects same-device COLLECTIVES with 'Duplicate GPU detected'
+        but serves same-device p2p fine — tools/nccl_probe.py)."""
+        if self.comm.world <= 1:
+            return
+        fp = self.flat.flat_param
+        if self.comm.is_ps:
+            for w, st in self.peers.items():
+                dist.isend(fp, dst=w, group=st.reply_g).wait()
+        else:
+            dist.irecv(fp, src=self.comm.ps_rank, group=self.reply_g).wait()
+
     # ------------------------------------------------------------------ PS
 
     def _start_ps(self):
