@@ -1,14 +1,21 @@
-"""Fused flash attention (ops/csrc/attn_kernels.hip).
+"""Fused flash attention (MFMA-tiled, ops/csrc/mfma_attn_kernels.hip).
 
-Correctness-first flash fwd+bwd, one wave per row, D == 64 — hardware-
-validated against torch SDPA (tests/test_gpu_attn.py) but not yet
-perf-competitive with aotriton, so the models keep torch SDPA.  ROADMAP.md
-item 2 is the MFMA-tiled rewrite of the inner loops.
+Two in-tree implementations, selected by PS_AMD_ATTN:
+  "mfma" (default) — 16x16x32-MFMA-tiled flash fwd+bwd, 128-row blocks,
+                     transposed-operand LDS staging (round 2);
+  "ref"            — the one-wave-per-row correctness kernels
+                     (attn_kernels.hip) — the on-GPU oracle;
+  "torch"          — torch SDPA (aotriton).
+
+Shape contract for the fused paths: [B, H, N, D] bf16 contiguous, D == 64
+(GPT-2-small and ViT-B/16 head dim).  Anything else falls back to torch
+SDPA.
 """
 
 from __future__ import annotations
 
 import math
+import os
 
 import torch
 import torch.nn.functional as F
@@ -16,16 +23,24 @@ import torch.nn.functional as F
 from . import HAVE_EXT, _EXT
 
 
+def _impl():
+    return os.environ.get("PS_AMD_ATTN", "mfma")
+
+
 class _FusedSDPA(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, causal):
+    def forward(ctx, q, k, v, causal, use_mfma):
         B, H, N, D = q.shape
         scale = 1.0 / math.sqrt(D)
         o = torch.empty_like(q)
         lse = torch.empty(B * H * N, dtype=torch.float32, device=q.device)
-        _EXT.attn_fwd(q, k, v, o, lse, N, scale, causal)
+        if use_mfma:
+            _EXT.fa_fwd(q, k, v, o, lse, N, scale, causal)
+        else:
+            _EXT.attn_fwd(q, k, v, o, lse, N, scale, causal)
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.causal = causal
+        ctx.use_mfma = use_mfma
         return o
 
     @staticmethod
@@ -38,16 +53,23 @@ class _FusedSDPA(torch.autograd.Function):
         dq = torch.empty_like(q)
         dk = torch.empty_like(k)
         dv = torch.empty_like(v)
-        _EXT.attn_bwd(q, k, v, o, dout, lse, delta, dq, dk, dv, N, scale,
-                      ctx.causal)
-        return dq, dk, dv, None
+        if ctx.use_mfma:
+            _EXT.fa_bwd(q, k, v, o, dout, lse, delta, dq, dk, dv, N, scale,
+                        ctx.causal)
+        else:
+            _EXT.attn_bwd(q, k, v, o, dout, lse, delta, dq, dk, dv, N, scale,
+                          ctx.causal)
+        return dq, dk, dv, None, None
 
 
 def fused_sdpa(q, k, v, is_causal=False):
     """Like F.scaled_dot_product_attention for [B,H,N,D] with D==64."""
-    if (HAVE_EXT and q.is_cuda and q.dtype == torch.bfloat16
+    impl = _impl()
+    if (impl in ("mfma", "ref") and HAVE_EXT and q.is_cuda
+            and q.dtype == torch.bfloat16
             and q.dim() == 4 and q.shape[-1] == 64
-            and q.is_contiguous() and k.is_contiguous()
-            and v.is_contiguous() and q.shape == k.shape == v.shape):
-        return _FusedSDPA.apply(q, k, v, bool(is_causal))
+            and q.shape == k.shape == v.shape):
+        return _FusedSDPA.apply(q.contiguous(), k.contiguous(),
+                                v.contiguous(), bool(is_causal),
+                                impl == "mfma")
     return F.scaled_dot_product_attention(q, k, v, is_causal=is_causal)

@@ -75,6 +75,14 @@ int ps_attn_bwd(void* stream, const void* q, const void* k, const void* v,
                 const void* o, const void* dout, const float* lse,
                 float* delta, void* dq, void* dk, void* dv, int64_t rows,
                 int64_t N, float scale, int causal);
+int ps_fa_fwd(void* stream, const void* q, const void* k, const void* v,
+              void* o, float* lse, int64_t BH, int64_t N, float scale,
+              int causal);
+int ps_fa_bwd(void* stream, const void* q, const void* k, const void* v,
+              const void* o, const void* dout, const float* lse, float* delta,
+              void* dq, void* dk, void* dv, int64_t BH, int64_t N,
+              float scale, int causal);
+int ps_fa_selfcheck(void* stream, const void* a, const void* b, float* c);
 }
 
 namespace {
@@ -467,9 +475,45 @@ void attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o,
            "attn_bwd");
 }
 
+void fa_fwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o,
+            at::Tensor lse, int64_t N, double scale, bool causal) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16 &&
+                  q.is_contiguous(),
+              "q must be contiguous bf16");
+  const int64_t BH = q.numel() / 64 / N;
+  throw_on(ps_fa_fwd(cur_stream(q), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                     o.data_ptr(), lse.data_ptr<float>(), BH, N, (float)scale,
+                     causal ? 1 : 0),
+           "fa_fwd");
+}
+
+void fa_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o,
+            at::Tensor dout, at::Tensor lse, at::Tensor delta, at::Tensor dq,
+            at::Tensor dk, at::Tensor dv, int64_t N, double scale,
+            bool causal) {
+  const int64_t BH = q.numel() / 64 / N;
+  throw_on(ps_fa_bwd(cur_stream(q), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                     o.data_ptr(), dout.data_ptr(), lse.data_ptr<float>(),
+                     delta.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
+                     dv.data_ptr(), BH, N, (float)scale, causal ? 1 : 0),
+           "fa_bwd");
+}
+
+void fa_selfcheck(at::Tensor a, at::Tensor b, at::Tensor c) {
+  TORCH_CHECK(a.numel() == 16 * 32 && b.numel() == 32 * 16 &&
+                  c.numel() == 16 * 16,
+              "selfcheck shapes: A[16,32] B[32,16] C[16,16]");
+  throw_on(ps_fa_selfcheck(cur_stream(a), a.data_ptr(), b.data_ptr(),
+                           c.data_ptr<float>()),
+           "fa_selfcheck");
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
+  m.def("fa_fwd", &fa_fwd, "MFMA-tiled flash attention forward");
+  m.def("fa_bwd", &fa_bwd, "MFMA-tiled flash attention backward");
+  m.def("fa_selfcheck", &fa_selfcheck, "16x16x32 MFMA fragment-layout check");
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);
   m.def("ln_fwd", &ln_fwd);

@@ -1,0 +1,621 @@
+// mfma_attn_kernels.hip — MFMA-tiled flash attention fwd+bwd for CDNA4
+// (gfx950 / MI355X).  Round-2 replacement of the one-wave-per-row
+// correctness kernels (attn_kernels.hip, kept as the on-GPU oracle).
+//
+// Shape contract: q,k,v,o,do,dq,dk,dv are [B,H,N,D] bf16 contiguous with
+// D == 64.  Same lse/delta conventions as the oracle kernels:
+//   lse[r] = m + log(l) on scale-applied scores; delta[r] = dO_r . O_r.
+//
+// Design (see /opt/skills/guides: MFMA §3, LDS §2, T2/T14 notes):
+//  * mfma_f32_16x16x32_bf16 everywhere; the contraction axis is always the
+//    head dim (d=64 -> 2 mfma) or a 32-wide tile (1 mfma).
+//  * fragment maps (verified on hardware by k_mfma_selfcheck + the numerics
+//    tests): A: row=lane&15, k=(lane>>4)*8+j; B: col=lane&15, same k;
+//    C/D: col=lane&15, row=(lane>>4)*4+reg.
+//  * "B-frag = 8 consecutive elements of a row" means any operand whose
+//    mfma-k axis is the head dim loads STRAIGHT from row-major HBM (Q, K,
+//    dO, V-as-dP-operand); only operands contracted over the 32-wide tile
+//    axis (V in PV, K in dQ, Q/dO in dK/dV) need a transposed LDS image,
+//    filled cooperatively once per tile and double-buffered (fwd) so there
+//    is one barrier per tile.
+//  * P / dS move from their C-layout registers to A-layout via a per-wave
+//    private LDS round trip (32x32 bf16, no cross-wave sync).
+//  * 4 waves x 32 rows = 128-row (fwd/dq) or 128-key (dkv) blocks; grid =
+//    (ceil(N/128), B*H); every wave owns its 32-row strip end to end.
+//  * masking: MASK = -1e30f scores (finite: exp stays exact-0 for real
+//    rows, no NaNs for fully-masked rows); causal tiles above the diagonal
+//    are skipped per wave; rows/keys >= N never store.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <cstdint>
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8v;
+typedef __attribute__((ext_vector_type(4))) float f32x4v;
+
+#define MFMA16(A, B, C) __builtin_amdgcn_mfma_f32_16x16x32_bf16(A, B, C, 0, 0, 0)
+
+#define FA_D 64
+#define FA_BM 128   // rows (fwd/dq) or keys (dkv) per 4-wave block
+#define FA_BK 32    // kv (fwd/dq) or q (dkv) tile
+#define FA_MASK -1e30f
+
+__device__ __forceinline__ bf16x8v fa_zero8() {
+  bf16x8v z;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) z[j] = (__bf16)0.0f;
+  return z;
+}
+
+// 8 consecutive bf16 of row `row` at element offset `off` (row-guarded).
+__device__ __forceinline__ bf16x8v fa_ldrow8(const __hip_bfloat16* base,
+                                             int64_t row, int64_t nrows,
+                                             int off) {
+  if (row < nrows) return *(const bf16x8v*)(base + row * FA_D + off);
+  return fa_zero8();
+}
+
+// cooperative transpose fill: src rows [r0, r0+32) x 64 cols -> dstT[64][32]
+// (dstT row stride 32 elems).  256 threads, 8 elems each.
+__device__ __forceinline__ void fa_fill_t(const __hip_bfloat16* src,
+                                          int64_t r0, int64_t nrows,
+                                          __hip_bfloat16* dstT) {
+  const int k = threadIdx.x >> 3;        // 0..31 source row in tile
+  const int d0 = (threadIdx.x & 7) * 8;  // 0..56
+  const bf16x8v v = fa_ldrow8(src, r0 + k, nrows, d0);
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    dstT[(d0 + j) * FA_BK + k] = (__hip_bfloat16)(float)v[j];
+}
+
+// ---------------------------------------------------------------------------
+// forward
+// ---------------------------------------------------------------------------
+
+__global__ void __launch_bounds__(256)
+k_fa_fwd(const __hip_bfloat16* __restrict__ q,
+         const __hip_bfloat16* __restrict__ k,
+         const __hip_bfloat16* __restrict__ v,
+         __hip_bfloat16* __restrict__ o, float* __restrict__ lse,
+         int64_t N, float scale, int causal) {
+  // lds: vt double buffer [2][64][32] + per-wave P [4][32][32]
+  __shared__ __align__(16) __hip_bfloat16 lds[2 * FA_D * FA_BK
+                                              + 4 * FA_BK * FA_BK];
+  const int lane = threadIdx.x & 63;
+  const int wv = threadIdx.x >> 6;
+  const int c = lane & 15;
+  const int g = lane >> 4;
+  const int64_t bh = blockIdx.y;
+  const int64_t m0 = (int64_t)blockIdx.x * FA_BM;
+  const int64_t row0 = m0 + wv * 32;
+  const __hip_bfloat16* qb = q + bh * N * FA_D;
+  const __hip_bfloat16* kb = k + bh * N * FA_D;
+  const __hip_bfloat16* vb = v + bh * N * FA_D;
+  __hip_bfloat16* pbuf = lds + 2 * FA_D * FA_BK + wv * FA_BK * FA_BK;
+
+  bf16x8v aQ[2][2];
+#pragma unroll
+  for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+      aQ[rf][kk] = fa_ldrow8(qb, row0 + rf * 16 + c, N, kk * 32 + g * 8);
+
+  float m[2][4], l[2][4];
+  f32x4v acc[2][4];
+#pragma unroll
+  for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      m[rf][r] = FA_MASK;
+      l[rf][r] = 0.0f;
+    }
+#pragma unroll
+  for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+    for (int df = 0; df < 4; ++df) acc[rf][df] = (f32x4v)(0.0f);
+
+  const int64_t kend = causal ? min(N, m0 + FA_BM) : N;
+  const int64_t ntiles = (kend + FA_BK - 1) / FA_BK;
+  if (ntiles <= 0) return;
+  fa_fill_t(vb, 0, N, lds);  // V tile 0 -> buffer 0
+  for (int64_t t = 0; t < ntiles; ++t) {
+    const int64_t k0 = t * FA_BK;
+    __syncthreads();  // vt[t&1] filled; prior tile's reads complete
+    if (t + 1 < ntiles)
+      fa_fill_t(vb, (t + 1) * FA_BK, N, lds + ((t + 1) & 1) * FA_D * FA_BK);
+    const __hip_bfloat16* vt = lds + (t & 1) * FA_D * FA_BK;
+    if (causal && k0 > row0 + 31) continue;  // above this wave's diagonal
+
+    // S = Q @ K^T  (B-frag: 8 consecutive d of key row -> direct load)
+    f32x4v S[2][2];
+#pragma unroll
+    for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+      for (int jf = 0; jf < 2; ++jf) S[rf][jf] = (f32x4v)(0.0f);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8v bK[2];
+#pragma unroll
+      for (int jf = 0; jf < 2; ++jf)
+        bK[jf] = fa_ldrow8(kb, k0 + jf * 16 + c, N, kk * 32 + g * 8);
+#pragma unroll
+      for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+        for (int jf = 0; jf < 2; ++jf)
+          S[rf][jf] = MFMA16(aQ[rf][kk], bK[jf], S[rf][jf]);
+    }
+
+    // scale + mask + online softmax
+    float mx[2][4];
+#pragma unroll
+    for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int64_t qrow = row0 + rf * 16 + g * 4 + r;
+        const int64_t cmax = causal ? min(N, qrow + 1) : N;
+#pragma unroll
+        for (int jf = 0; jf < 2; ++jf) {
+          float s = S[rf][jf][r] * scale;
+          if (k0 + jf * 16 + c >= cmax) s = FA_MASK;
+          S[rf][jf][r] = s;
+        }
+        float t2 = fmaxf(S[rf][0][r], S[rf][1][r]);
+#pragma unroll
+        for (int d = 1; d < 16; d <<= 1) t2 = fmaxf(t2, __shfl_xor(t2, d, 16));
+        mx[rf][r] = t2;
+      }
+#pragma unroll
+    for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const float mn = fmaxf(m[rf][r], mx[rf][r]);
+        const float al = __expf(m[rf][r] - mn);
+        m[rf][r] = mn;
+        l[rf][r] *= al;
+#pragma unroll
+        for (int df = 0; df < 4; ++df) acc[rf][df][r] *= al;
+        float rs = 0.0f;
+#pragma unroll
+        for (int jf = 0; jf < 2; ++jf) {
+          const float p = __expf(S[rf][jf][r] - mn);
+          S[rf][jf][r] = p;
+          rs += p;
+        }
+#pragma unroll
+        for (int d = 1; d < 16; d <<= 1) rs += __shfl_xor(rs, d, 16);
+        l[rf][r] += rs;
+      }
+
+    // P (C-layout) -> per-wave LDS -> A-frags
+#pragma unroll
+    for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+      for (int jf = 0; jf < 2; ++jf)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          pbuf[(rf * 16 + g * 4 + r) * FA_BK + jf * 16 + c] =
+              (__hip_bfloat16)S[rf][jf][r];
+    bf16x8v aP[2];
+#pragma unroll
+    for (int rf = 0; rf < 2; ++rf)
+      aP[rf] = *(const bf16x8v*)(pbuf + (rf * 16 + c) * FA_BK + g * 8);
+
+    // O += P @ V   (B-frag from transposed V image)
+#pragma unroll
+    for (int df = 0; df < 4; ++df) {
+      const bf16x8v bV =
+          *(const bf16x8v*)(vt + (df * 16 + c) * FA_BK + g * 8);
+#pragma unroll
+      for (int rf = 0; rf < 2; ++rf)
+        acc[rf][df] = MFMA16(aP[rf], bV, acc[rf][df]);
+    }
+  }
+
+  __hip_bfloat16* ob = o + bh * N * FA_D;
+#pragma unroll
+  for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int64_t row = row0 + rf * 16 + g * 4 + r;
+      if (row >= N) continue;
+      const float inv = 1.0f / l[rf][r];
+#pragma unroll
+      for (int df = 0; df < 4; ++df)
+        ob[row * FA_D + df * 16 + c] =
+            (__hip_bfloat16)(acc[rf][df][r] * inv);
+      if (c == 0) lse[bh * N + row] = m[rf][r] + __logf(l[rf][r]);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// backward: delta then dq then dkv
+// ---------------------------------------------------------------------------
+
+__global__ void __launch_bounds__(256)
+k_fa_delta(const __hip_bfloat16* __restrict__ dout,
+           const __hip_bfloat16* __restrict__ o, float* __restrict__ delta,
+           int64_t rows) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int64_t rstride = (int64_t)gridDim.x * 4;
+  for (int64_t r = (int64_t)blockIdx.x * 4 + wave; r < rows; r += rstride) {
+    float d = (float)dout[r * FA_D + lane] * (float)o[r * FA_D + lane];
+#pragma unroll
+    for (int s = 1; s < 64; s <<= 1) d += __shfl_xor(d, s, 64);
+    if (lane == 0) delta[r] = d;
+  }
+}
+
+__global__ void __launch_bounds__(256)
+k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
+            const __hip_bfloat16* __restrict__ k,
+            const __hip_bfloat16* __restrict__ v,
+            const __hip_bfloat16* __restrict__ dout,
+            const float* __restrict__ lse, const float* __restrict__ delta,
+            __hip_bfloat16* __restrict__ dq, int64_t N, float scale,
+            int causal) {
+  // lds: Kt double buffer [2][64][32] + per-wave dS [4][32][32]
+  __shared__ __align__(16) __hip_bfloat16 lds[2 * FA_D * FA_BK
+                                              + 4 * FA_BK * FA_BK];
+  const int lane = threadIdx.x & 63;
+  const int wv = threadIdx.x >> 6;
+  const int c = lane & 15;
+  const int g = lane >> 4;
+  const int64_t bh = blockIdx.y;
+  const int64_t m0 = (int64_t)blockIdx.x * FA_BM;
+  const int64_t row0 = m0 + wv * 32;
+  const __hip_bfloat16* qb = q + bh * N * FA_D;
+  const __hip_bfloat16* kb = k + bh * N * FA_D;
+  const __hip_bfloat16* vb = v + bh * N * FA_D;
+  const __hip_bfloat16* dob = dout + bh * N * FA_D;
+  __hip_bfloat16* sbuf = lds + 2 * FA_D * FA_BK + wv * FA_BK * FA_BK;
+
+  bf16x8v aQ[2][2], aDO[2][2];
+  float lse_r[2][4], dl_r[2][4];
+#pragma unroll
+  for (int rf = 0; rf < 2; ++rf) {
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      aQ[rf][kk] = fa_ldrow8(qb, row0 + rf * 16 + c, N, kk * 32 + g * 8);
+      aDO[rf][kk] = fa_ldrow8(dob, row0 + rf * 16 + c, N, kk * 32 + g * 8);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int64_t row = row0 + rf * 16 + g * 4 + r;
+      lse_r[rf][r] = (row < N) ? lse[bh * N + row] : 0.0f;
+      dl_r[rf][r] = (row < N) ? delta[bh * N + row] : 0.0f;
+    }
+  }
+  f32x4v acc[2][4];
+#pragma unroll
+  for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+    for (int df = 0; df < 4; ++df) acc[rf][df] = (f32x4v)(0.0f);
+
+  const int64_t kend = causal ? min(N, m0 + FA_BM) : N;
+  const int64_t ntiles = (kend + FA_BK - 1) / FA_BK;
+  if (ntiles <= 0) return;
+  fa_fill_t(kb, 0, N, lds);
+  for (int64_t t = 0; t < ntiles; ++t) {
+    const int64_t k0 = t * FA_BK;
+    __syncthreads();
+    if (t + 1 < ntiles)
+      fa_fill_t(kb, (t + 1) * FA_BK, N, lds + ((t + 1) & 1) * FA_D * FA_BK);
+    const __hip_bfloat16* kt = lds + (t & 1) * FA_D * FA_BK;
+    if (causal && k0 > row0 + 31) continue;
+
+    // S and dP in one pass over kk
+    f32x4v S[2][2], dP[2][2];
+#pragma unroll
+    for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+      for (int jf = 0; jf < 2; ++jf) {
+        S[rf][jf] = (f32x4v)(0.0f);
+        dP[rf][jf] = (f32x4v)(0.0f);
+      }
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+      for (int jf = 0; jf < 2; ++jf) {
+        const bf16x8v bK = fa_ldrow8(kb, k0 + jf * 16 + c, N,
+                                     kk * 32 + g * 8);
+        const bf16x8v bV = fa_ldrow8(vb, k0 + jf * 16 + c, N,
+                                     kk * 32 + g * 8);
+#pragma unroll
+        for (int rf = 0; rf < 2; ++rf) {
+          S[rf][jf] = MFMA16(aQ[rf][kk], bK, S[rf][jf]);
+          dP[rf][jf] = MFMA16(aDO[rf][kk], bV, dP[rf][jf]);
+        }
+      }
+    }
+    // dS = P * (dP - delta)
+#pragma unroll
+    for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int64_t qrow = row0 + rf * 16 + g * 4 + r;
+        const int64_t cmax = causal ? min(N, qrow + 1) : N;
+#pragma unroll
+        for (int jf = 0; jf < 2; ++jf) {
+          float s = S[rf][jf][r] * scale;
+          if (k0 + jf * 16 + c >= cmax) s = FA_MASK;
+          const float p = __expf(s - lse_r[rf][r]);
+          S[rf][jf][r] = p * (dP[rf][jf][r] - dl_r[rf][r]);
+        }
+      }
+    // dS -> LDS -> A-frags; dQ += dS @ K (B-frag from transposed K image)
+#pragma unroll
+    for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+      for (int jf = 0; jf < 2; ++jf)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          sbuf[(rf * 16 + g * 4 + r) * FA_BK + jf * 16 + c] =
+              (__hip_bfloat16)S[rf][jf][r];
+    bf16x8v aDS[2];
+#pragma unroll
+    for (int rf = 0; rf < 2; ++rf)
+      aDS[rf] = *(const bf16x8v*)(sbuf + (rf * 16 + c) * FA_BK + g * 8);
+#pragma unroll
+    for (int df = 0; df < 4; ++df) {
+      const bf16x8v bKt =
+          *(const bf16x8v*)(kt + (df * 16 + c) * FA_BK + g * 8);
+#pragma unroll
+      for (int rf = 0; rf < 2; ++rf)
+        acc[rf][df] = MFMA16(aDS[rf], bKt, acc[rf][df]);
+    }
+  }
+
+  __hip_bfloat16* dqb = dq + bh * N * FA_D;
+#pragma unroll
+  for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int64_t row = row0 + rf * 16 + g * 4 + r;
+      if (row >= N) continue;
+#pragma unroll
+      for (int df = 0; df < 4; ++df)
+        dqb[row * FA_D + df * 16 + c] =
+            (__hip_bfloat16)(acc[rf][df][r] * scale);
+    }
+}
+
+__global__ void __launch_bounds__(256)
+k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
+             const __hip_bfloat16* __restrict__ k,
+             const __hip_bfloat16* __restrict__ v,
+             const __hip_bfloat16* __restrict__ dout,
+             const float* __restrict__ lse, const float* __restrict__ delta,
+             __hip_bfloat16* __restrict__ dk, __hip_bfloat16* __restrict__ dv,
+             int64_t N, float scale, int causal) {
+  // lds: Qt [64][32] + dOt [64][32] + per-wave P/dS [4][32][32]
+  __shared__ __align__(16) __hip_bfloat16 lds[2 * FA_D * FA_BK
+                                              + 4 * FA_BK * FA_BK];
+  const int lane = threadIdx.x & 63;
+  const int wv = threadIdx.x >> 6;
+  const int c = lane & 15;
+  const int g = lane >> 4;
+  const int64_t bh = blockIdx.y;
+  const int64_t m0 = (int64_t)blockIdx.x * FA_BM;  // first KEY of block
+  const int64_t key0 = m0 + wv * 32;               // wave's first key
+  const __hip_bfloat16* qb = q + bh * N * FA_D;
+  const __hip_bfloat16* kb = k + bh * N * FA_D;
+  const __hip_bfloat16* vb = v + bh * N * FA_D;
+  const __hip_bfloat16* dob = dout + bh * N * FA_D;
+  __hip_bfloat16* qt = lds;                       // [64][32]
+  __hip_bfloat16* dot = lds + FA_D * FA_BK;       // [64][32]
+  __hip_bfloat16* sbuf = lds + 2 * FA_D * FA_BK + wv * FA_BK * FA_BK;
+
+  bf16x8v aK[2][2], aV[2][2];
+#pragma unroll
+  for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      aK[rf][kk] = fa_ldrow8(kb, key0 + rf * 16 + c, N, kk * 32 + g * 8);
+      aV[rf][kk] = fa_ldrow8(vb, key0 + rf * 16 + c, N, kk * 32 + g * 8);
+    }
+  f32x4v dK[2][4], dV[2][4];
+#pragma unroll
+  for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+    for (int df = 0; df < 4; ++df) {
+      dK[rf][df] = (f32x4v)(0.0f);
+      dV[rf][df] = (f32x4v)(0.0f);
+    }
+
+  const int64_t q0_start = causal ? m0 : 0;
+  for (int64_t q0 = q0_start; q0 < N; q0 += FA_BK) {
+    __syncthreads();  // prior tile's Qt/dOt reads complete
+    fa_fill_t(qb, q0, N, qt);
+    fa_fill_t(dob, q0, N, dot);
+    __syncthreads();
+    if (causal && q0 + 31 < key0) continue;  // entirely below diagonal
+
+    float lse_c[2], dl_c[2];
+#pragma unroll
+    for (int jf = 0; jf < 2; ++jf) {
+      const int64_t qcol = q0 + jf * 16 + c;
+      lse_c[jf] = (qcol < N) ? lse[bh * N + qcol] : 0.0f;
+      dl_c[jf] = (qcol < N) ? delta[bh * N + qcol] : 0.0f;
+    }
+
+    // S^T = K @ Q^T and dP^T = V @ dO^T (B-frags direct from rows)
+    f32x4v St[2][2], dPt[2][2];
+#pragma unroll
+    for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+      for (int jf = 0; jf < 2; ++jf) {
+        St[rf][jf] = (f32x4v)(0.0f);
+        dPt[rf][jf] = (f32x4v)(0.0f);
+      }
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+      for (int jf = 0; jf < 2; ++jf) {
+        const bf16x8v bQ = fa_ldrow8(qb, q0 + jf * 16 + c, N,
+                                     kk * 32 + g * 8);
+        const bf16x8v bDO = fa_ldrow8(dob, q0 + jf * 16 + c, N,
+                                      kk * 32 + g * 8);
+#pragma unroll
+        for (int rf = 0; rf < 2; ++rf) {
+          St[rf][jf] = MFMA16(aK[rf][kk], bQ, St[rf][jf]);
+          dPt[rf][jf] = MFMA16(aV[rf][kk], bDO, dPt[rf][jf]);
+        }
+      }
+    }
+
+    // P^T = exp(s*scale - lse[q]);  dS^T = P^T * (dP^T - delta[q])
+    float Pt[2][2][4];
+#pragma unroll
+    for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int64_t keyrow = key0 + rf * 16 + g * 4 + r;
+#pragma unroll
+        for (int jf = 0; jf < 2; ++jf) {
+          const int64_t qcol = q0 + jf * 16 + c;
+          float s = St[rf][jf][r] * scale;
+          if (qcol >= N || (causal && qcol < keyrow)) s = FA_MASK;
+          const float p = __expf(s - lse_c[jf]);
+          Pt[rf][jf][r] = p;
+          St[rf][jf][r] = p * (dPt[rf][jf][r] - dl_c[jf]);
+        }
+      }
+
+    // P^T -> LDS -> A-frags; dV += P^T @ dO (B from dOt image)
+#pragma unroll
+    for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+      for (int jf = 0; jf < 2; ++jf)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          sbuf[(rf * 16 + g * 4 + r) * FA_BK + jf * 16 + c] =
+              (__hip_bfloat16)Pt[rf][jf][r];
+    bf16x8v aPT[2];
+#pragma unroll
+    for (int rf = 0; rf < 2; ++rf)
+      aPT[rf] = *(const bf16x8v*)(sbuf + (rf * 16 + c) * FA_BK + g * 8);
+#pragma unroll
+    for (int df = 0; df < 4; ++df) {
+      const bf16x8v bDOt =
+          *(const bf16x8v*)(dot + (df * 16 + c) * FA_BK + g * 8);
+#pragma unroll
+      for (int rf = 0; rf < 2; ++rf)
+        dV[rf][df] = MFMA16(aPT[rf], bDOt, dV[rf][df]);
+    }
+
+    // dS^T -> LDS -> A-frags; dK += dS^T @ Q (B from Qt image)
+#pragma unroll
+    for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+      for (int jf = 0; jf < 2; ++jf)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          sbuf[(rf * 16 + g * 4 + r) * FA_BK + jf * 16 + c] =
+              (__hip_bfloat16)St[rf][jf][r];
+    bf16x8v aDST[2];
+#pragma unroll
+    for (int rf = 0; rf < 2; ++rf)
+      aDST[rf] = *(const bf16x8v*)(sbuf + (rf * 16 + c) * FA_BK + g * 8);
+#pragma unroll
+    for (int df = 0; df < 4; ++df) {
+      const bf16x8v bQt =
+          *(const bf16x8v*)(qt + (df * 16 + c) * FA_BK + g * 8);
+#pragma unroll
+      for (int rf = 0; rf < 2; ++rf)
+        dK[rf][df] = MFMA16(aDST[rf], bQt, dK[rf][df]);
+    }
+  }
+
+  __hip_bfloat16* dkb = dk + bh * N * FA_D;
+  __hip_bfloat16* dvb = dv + bh * N * FA_D;
+#pragma unroll
+  for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int64_t row = key0 + rf * 16 + g * 4 + r;
+      if (row >= N) continue;
+#pragma unroll
+      for (int df = 0; df < 4; ++df) {
+        dkb[row * FA_D + df * 16 + c] =
+            (__hip_bfloat16)(dK[rf][df][r] * scale);
+        dvb[row * FA_D + df * 16 + c] = (__hip_bfloat16)dV[rf][df][r];
+      }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// fragment-layout self-check: one wave computes C[16][16] = A[16][32] @
+// B^T? no — B given as [32][16] column-slices per the assumed B layout.
+// Host compares against a reference matmul; a layout mismatch fails loudly.
+// ---------------------------------------------------------------------------
+
+__global__ void k_fa_selfcheck(const __hip_bfloat16* __restrict__ A,
+                               const __hip_bfloat16* __restrict__ B,
+                               float* __restrict__ C) {
+  if (threadIdx.x >= 64) return;
+  const int lane = threadIdx.x;
+  const int c = lane & 15;
+  const int g = lane >> 4;
+  bf16x8v a, b;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    a[j] = A[c * 32 + g * 8 + j];      // A[row=c][k]
+    b[j] = B[(g * 8 + j) * 16 + c];    // B[k][col=c]
+  }
+  f32x4v acc = (f32x4v)(0.0f);
+  acc = MFMA16(a, b, acc);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) C[(g * 4 + r) * 16 + c] = acc[r];
+}
+
+// ---------------------------------------------------------------------------
+// C ABI launchers
+// ---------------------------------------------------------------------------
+
+extern "C" {
+
+int ps_fa_fwd(void* stream_, const void* q, const void* k, const void* v,
+              void* o, float* lse, int64_t BH, int64_t N, float scale,
+              int causal) {
+  hipStream_t s = (hipStream_t)stream_;
+  dim3 grid((unsigned)((N + FA_BM - 1) / FA_BM), (unsigned)BH);
+  hipLaunchKernelGGL(k_fa_fwd, grid, dim3(256), 0, s,
+                     (const __hip_bfloat16*)q, (const __hip_bfloat16*)k,
+                     (const __hip_bfloat16*)v, (__hip_bfloat16*)o, lse, N,
+                     scale, causal);
+  return (int)hipGetLastError();
+}
+
+int ps_fa_bwd(void* stream_, const void* q, const void* k, const void* v,
+              const void* o, const void* dout, const float* lse, float* delta,
+              void* dq, void* dk, void* dv, int64_t BH, int64_t N,
+              float scale, int causal) {
+  hipStream_t s = (hipStream_t)stream_;
+  const int64_t rows = BH * N;
+  int64_t db = (rows + 3) / 4;
+  if (db > 2048) db = 2048;
+  hipLaunchKernelGGL(k_fa_delta, dim3((unsigned)db), dim3(256), 0, s,
+                     (const __hip_bfloat16*)dout, (const __hip_bfloat16*)o,
+                     delta, rows);
+  dim3 grid((unsigned)((N + FA_BM - 1) / FA_BM), (unsigned)BH);
+  hipLaunchKernelGGL(k_fa_bwd_dq, grid, dim3(256), 0, s,
+                     (const __hip_bfloat16*)q, (const __hip_bfloat16*)k,
+                     (const __hip_bfloat16*)v, (const __hip_bfloat16*)dout,
+                     lse, delta, (__hip_bfloat16*)dq, N, scale, causal);
+  hipLaunchKernelGGL(k_fa_bwd_dkv, grid, dim3(256), 0, s,
+                     (const __hip_bfloat16*)q, (const __hip_bfloat16*)k,
+                     (const __hip_bfloat16*)v, (const __hip_bfloat16*)dout,
+                     lse, delta, (__hip_bfloat16*)dk, (__hip_bfloat16*)dv, N,
+                     scale, causal);
+  return (int)hipGetLastError();
+}
+
+int ps_fa_selfcheck(void* stream_, const void* a, const void* b, float* c) {
+  hipStream_t s = (hipStream_t)stream_;
+  hipLaunchKernelGGL(k_fa_selfcheck, dim3(1), dim3(64), 0, s,
+                     (const __hip_bfloat16*)a, (const __hip_bfloat16*)b, c);
+  return (int)hipGetLastError();
+}
+
+}  // extern "C"

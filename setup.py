@@ -27,6 +27,7 @@ setup(
                 os.path.join(CSRC, "ln_kernels.hip"),
                 os.path.join(CSRC, "ce_kernels.hip"),
                 os.path.join(CSRC, "attn_kernels.hip"),
+                os.path.join(CSRC, "mfma_attn_kernels.hip"),
             ],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],

@@ -1,23 +1,46 @@
 """Fused flash-attention fwd+bwd vs torch SDPA (@gpu).
 
-Hardware-validated (causal + non-causal, odd N, all three grads); the
-MFMA-tiled perf rewrite is ROADMAP item 2 — models keep torch SDPA until
-this kernel beats it.
+Covers BOTH in-tree implementations (PS_AMD_ATTN=mfma|ref) against the
+fp32 torch oracle, at the model shapes (GPT-2 N=512 causal, ViT N=197
+non-causal) and edge shapes (odd N, tiny N), plus the 16x16x32 MFMA
+fragment-layout self-check that pins the A/B/C lane maps the MFMA kernels
+assume.
 """
 
 import pytest
 import torch
 import torch.nn.functional as F
 
+from pytorch_ps_mpi_amd import ops
 from pytorch_ps_mpi_amd.ops.attn import fused_sdpa
 
 pytestmark = pytest.mark.gpu
 
 
+def test_mfma_fragment_layouts():
+    """C[16,16] = A[16,32] @ B[32,16] through one mfma_f32_16x16x32_bf16
+    with the lane maps the attention kernels assume.  Asymmetric operands
+    (catches transposed layouts, guide §5.4 rule 16)."""
+    dev = "cuda:0"
+    torch.manual_seed(0)
+    a = torch.randn(16, 32, device=dev).bfloat16()
+    b = torch.randn(32, 16, device=dev).bfloat16()
+    c = torch.zeros(16, 16, dtype=torch.float32, device=dev)
+    ops._EXT.fa_selfcheck(a.contiguous(), b.contiguous(), c)
+    torch.cuda.synchronize()
+    ref = a.float() @ b.float()
+    err = (c - ref).abs().max().item()
+    assert err < 0.1, f"fragment layout mismatch: max err {err}"
+
+
+@pytest.mark.parametrize("impl", ["mfma", "ref"])
 @pytest.mark.parametrize("B,H,N,causal", [
     (2, 2, 64, False), (2, 2, 67, False), (1, 3, 128, True), (2, 1, 33, True),
+    (2, 2, 197, False),   # ViT-B/16 sequence (odd, multi-block)
+    (1, 2, 512, True),    # GPT-2 sequence
 ])
-def test_fused_sdpa_fwd_bwd(B, H, N, causal):
+def test_fused_sdpa_fwd_bwd(impl, B, H, N, causal, monkeypatch):
+    monkeypatch.setenv("PS_AMD_ATTN", impl)
     torch.manual_seed(0)
     dev = "cuda:0"
     D = 64
@@ -42,6 +65,22 @@ def test_fused_sdpa_fwd_bwd(B, H, N, causal):
         err = (got.float() - want).abs().max().item()
         scale = want.abs().max().item() + 1.0
         assert err < 0.05 * scale, f"{name} err {err}"
+
+
+def test_mfma_vs_ref_spike_row(monkeypatch):
+    """A spiked K row forces large online-softmax rescales mid-sequence
+    (guide §5.4 rule 26: bounded random data never exercises that branch)."""
+    monkeypatch.setenv("PS_AMD_ATTN", "mfma")
+    torch.manual_seed(1)
+    dev = "cuda:0"
+    B, H, N, D = 1, 2, 160, 64
+    q32 = torch.randn(B, H, N, D, device=dev).bfloat16().float()
+    k32 = torch.randn(B, H, N, D, device=dev).bfloat16().float()
+    k32[:, :, 140] *= 8.0  # spike near the end -> max jumps at a late tile
+    v32 = torch.randn(B, H, N, D, device=dev).bfloat16().float()
+    ref = F.scaled_dot_product_attention(q32, k32, v32)
+    o = fused_sdpa(q32.bfloat16(), k32.bfloat16(), v32.bfloat16())
+    assert (o.float() - ref).abs().max().item() < 0.06
 
 
 def test_fused_sdpa_fallback_d128():
