@@ -8,18 +8,21 @@
 //
 // Design (see /opt/skills/guides: MFMA §3, LDS §2, T2/T14 notes):
 //  * mfma_f32_16x16x32_bf16 everywhere; the contraction axis is always the
-//    head dim (d=64 -> 2 mfma) or a 32-wide tile (1 mfma).
-//  * fragment maps (verified on hardware by k_mfma_selfcheck + the numerics
+//    head dim (d=64 -> 2 mfma) or the 64-wide tile axis (2 mfma).
+//  * fragment maps (verified on hardware by k_fa_selfcheck + numerics
 //    tests): A: row=lane&15, k=(lane>>4)*8+j; B: col=lane&15, same k;
 //    C/D: col=lane&15, row=(lane>>4)*4+reg.
 //  * "B-frag = 8 consecutive elements of a row" means any operand whose
 //    mfma-k axis is the head dim loads STRAIGHT from row-major HBM (Q, K,
-//    dO, V-as-dP-operand); only operands contracted over the 32-wide tile
-//    axis (V in PV, K in dQ, Q/dO in dK/dV) need a transposed LDS image,
-//    filled cooperatively once per tile and double-buffered (fwd) so there
-//    is one barrier per tile.
+//    dO, V-as-dP-operand); only operands contracted over the tile axis
+//    (V in PV, K in dQ, Q/dO in dK/dV) need a transposed LDS image, filled
+//    cooperatively once per tile (double-buffered in fwd/dq: one barrier
+//    per tile).
 //  * P / dS move from their C-layout registers to A-layout via a per-wave
-//    private LDS round trip (32x32 bf16, no cross-wave sync).
+//    private LDS round trip (32x64 bf16, no cross-wave sync).
+//  * 64-key (fwd/dq) / 64-query (dkv) tiles: 32 MFMAs per wave per tile,
+//    which halves the softmax-VALU + LDS-roundtrip cost per key vs a
+//    32-wide tile (measured 77 -> see profiles/attn_bench).
 //  * 4 waves x 32 rows = 128-row (fwd/dq) or 128-key (dkv) blocks; grid =
 //    (ceil(N/128), B*H); every wave owns its 32-row strip end to end.
 //  * masking: MASK = -1e30f scores (finite: exp stays exact-0 for real
@@ -37,7 +40,7 @@ typedef __attribute__((ext_vector_type(4))) float f32x4v;
 
 #define FA_D 64
 #define FA_BM 128   // rows (fwd/dq) or keys (dkv) per 4-wave block
-#define FA_BK 32    // kv (fwd/dq) or q (dkv) tile
+#define FA_BN 64    // kv (fwd/dq) or q (dkv) tile width
 #define FA_MASK -1e30f
 
 __device__ __forceinline__ bf16x8v fa_zero8() {
@@ -55,17 +58,20 @@ __device__ __forceinline__ bf16x8v fa_ldrow8(const __hip_bfloat16* base,
   return fa_zero8();
 }
 
-// cooperative transpose fill: src rows [r0, r0+32) x 64 cols -> dstT[64][32]
-// (dstT row stride 32 elems).  256 threads, 8 elems each.
+// cooperative transpose fill: src rows [r0, r0+64) x 64 cols -> dstT[64][64]
+// (dstT row stride FA_BN).  256 threads, 16 elems each.
 __device__ __forceinline__ void fa_fill_t(const __hip_bfloat16* src,
                                           int64_t r0, int64_t nrows,
                                           __hip_bfloat16* dstT) {
-  const int k = threadIdx.x >> 3;        // 0..31 source row in tile
-  const int d0 = (threadIdx.x & 7) * 8;  // 0..56
-  const bf16x8v v = fa_ldrow8(src, r0 + k, nrows, d0);
+  const int k = threadIdx.x >> 2;         // 0..63 source row in tile
+  const int d0 = (threadIdx.x & 3) * 16;  // 0..48
 #pragma unroll
-  for (int j = 0; j < 8; ++j)
-    dstT[(d0 + j) * FA_BK + k] = (__hip_bfloat16)(float)v[j];
+  for (int h = 0; h < 2; ++h) {
+    const bf16x8v v = fa_ldrow8(src, r0 + k, nrows, d0 + h * 8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      dstT[(d0 + h * 8 + j) * FA_BN + k] = (__hip_bfloat16)(float)v[j];
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -78,9 +84,9 @@ k_fa_fwd(const __hip_bfloat16* __restrict__ q,
          const __hip_bfloat16* __restrict__ v,
          __hip_bfloat16* __restrict__ o, float* __restrict__ lse,
          int64_t N, float scale, int causal) {
-  // lds: vt double buffer [2][64][32] + per-wave P [4][32][32]
-  __shared__ __align__(16) __hip_bfloat16 lds[2 * FA_D * FA_BK
-                                              + 4 * FA_BK * FA_BK];
+  // lds: vt double buffer [2][64][64] + per-wave P [4][32][64]
+  __shared__ __align__(16) __hip_bfloat16 lds[2 * FA_D * FA_BN
+                                              + 4 * 32 * FA_BN];
   const int lane = threadIdx.x & 63;
   const int wv = threadIdx.x >> 6;
   const int c = lane & 15;
@@ -91,7 +97,7 @@ k_fa_fwd(const __hip_bfloat16* __restrict__ q,
   const __hip_bfloat16* qb = q + bh * N * FA_D;
   const __hip_bfloat16* kb = k + bh * N * FA_D;
   const __hip_bfloat16* vb = v + bh * N * FA_D;
-  __hip_bfloat16* pbuf = lds + 2 * FA_D * FA_BK + wv * FA_BK * FA_BK;
+  __hip_bfloat16* pbuf = lds + 2 * FA_D * FA_BN + wv * 32 * FA_BN;
 
   bf16x8v aQ[2][2];
 #pragma unroll
@@ -115,38 +121,36 @@ k_fa_fwd(const __hip_bfloat16* __restrict__ q,
     for (int df = 0; df < 4; ++df) acc[rf][df] = (f32x4v)(0.0f);
 
   const int64_t kend = causal ? min(N, m0 + FA_BM) : N;
-  const int64_t ntiles = (kend + FA_BK - 1) / FA_BK;
+  const int64_t ntiles = (kend + FA_BN - 1) / FA_BN;
   if (ntiles <= 0) return;
   fa_fill_t(vb, 0, N, lds);  // V tile 0 -> buffer 0
   for (int64_t t = 0; t < ntiles; ++t) {
-    const int64_t k0 = t * FA_BK;
+    const int64_t k0 = t * FA_BN;
     __syncthreads();  // vt[t&1] filled; prior tile's reads complete
     if (t + 1 < ntiles)
-      fa_fill_t(vb, (t + 1) * FA_BK, N, lds + ((t + 1) & 1) * FA_D * FA_BK);
-    const __hip_bfloat16* vt = lds + (t & 1) * FA_D * FA_BK;
+      fa_fill_t(vb, (t + 1) * FA_BN, N, lds + ((t + 1) & 1) * FA_D * FA_BN);
+    const __hip_bfloat16* vt = lds + (t & 1) * FA_D * FA_BN;
     if (causal && k0 > row0 + 31) continue;  // above this wave's diagonal
 
     // S = Q @ K^T  (B-frag: 8 consecutive d of key row -> direct load)
-    f32x4v S[2][2];
+    f32x4v S[2][4];
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
 #pragma unroll
-      for (int jf = 0; jf < 2; ++jf) S[rf][jf] = (f32x4v)(0.0f);
+      for (int jf = 0; jf < 4; ++jf) S[rf][jf] = (f32x4v)(0.0f);
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      bf16x8v bK[2];
 #pragma unroll
-      for (int jf = 0; jf < 2; ++jf)
-        bK[jf] = fa_ldrow8(kb, k0 + jf * 16 + c, N, kk * 32 + g * 8);
+      for (int jf = 0; jf < 4; ++jf) {
+        const bf16x8v bK = fa_ldrow8(kb, k0 + jf * 16 + c, N,
+                                     kk * 32 + g * 8);
 #pragma unroll
-      for (int rf = 0; rf < 2; ++rf)
-#pragma unroll
-        for (int jf = 0; jf < 2; ++jf)
-          S[rf][jf] = MFMA16(aQ[rf][kk], bK[jf], S[rf][jf]);
+        for (int rf = 0; rf < 2; ++rf)
+          S[rf][jf] = MFMA16(aQ[rf][kk], bK, S[rf][jf]);
+      }
     }
 
     // scale + mask + online softmax
-    float mx[2][4];
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
 #pragma unroll
@@ -154,21 +158,16 @@ k_fa_fwd(const __hip_bfloat16* __restrict__ q,
         const int64_t qrow = row0 + rf * 16 + g * 4 + r;
         const int64_t cmax = causal ? min(N, qrow + 1) : N;
 #pragma unroll
-        for (int jf = 0; jf < 2; ++jf) {
+        for (int jf = 0; jf < 4; ++jf) {
           float s = S[rf][jf][r] * scale;
           if (k0 + jf * 16 + c >= cmax) s = FA_MASK;
           S[rf][jf][r] = s;
         }
-        float t2 = fmaxf(S[rf][0][r], S[rf][1][r]);
+        float t2 = fmaxf(fmaxf(S[rf][0][r], S[rf][1][r]),
+                         fmaxf(S[rf][2][r], S[rf][3][r]));
 #pragma unroll
         for (int d = 1; d < 16; d <<= 1) t2 = fmaxf(t2, __shfl_xor(t2, d, 16));
-        mx[rf][r] = t2;
-      }
-#pragma unroll
-    for (int rf = 0; rf < 2; ++rf)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const float mn = fmaxf(m[rf][r], mx[rf][r]);
+        const float mn = fmaxf(m[rf][r], t2);
         const float al = __expf(m[rf][r] - mn);
         m[rf][r] = mn;
         l[rf][r] *= al;
@@ -176,7 +175,7 @@ k_fa_fwd(const __hip_bfloat16* __restrict__ q,
         for (int df = 0; df < 4; ++df) acc[rf][df][r] *= al;
         float rs = 0.0f;
 #pragma unroll
-        for (int jf = 0; jf < 2; ++jf) {
+        for (int jf = 0; jf < 4; ++jf) {
           const float p = __expf(S[rf][jf][r] - mn);
           S[rf][jf][r] = p;
           rs += p;
@@ -190,25 +189,30 @@ k_fa_fwd(const __hip_bfloat16* __restrict__ q,
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
 #pragma unroll
-      for (int jf = 0; jf < 2; ++jf)
+      for (int jf = 0; jf < 4; ++jf)
 #pragma unroll
         for (int r = 0; r < 4; ++r)
-          pbuf[(rf * 16 + g * 4 + r) * FA_BK + jf * 16 + c] =
+          pbuf[(rf * 16 + g * 4 + r) * FA_BN + jf * 16 + c] =
               (__hip_bfloat16)S[rf][jf][r];
-    bf16x8v aP[2];
+    bf16x8v aP[2][2];
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
-      aP[rf] = *(const bf16x8v*)(pbuf + (rf * 16 + c) * FA_BK + g * 8);
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+        aP[rf][kk] = *(const bf16x8v*)(pbuf + (rf * 16 + c) * FA_BN
+                                       + kk * 32 + g * 8);
 
     // O += P @ V   (B-frag from transposed V image)
 #pragma unroll
-    for (int df = 0; df < 4; ++df) {
-      const bf16x8v bV =
-          *(const bf16x8v*)(vt + (df * 16 + c) * FA_BK + g * 8);
+    for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
-      for (int rf = 0; rf < 2; ++rf)
-        acc[rf][df] = MFMA16(aP[rf], bV, acc[rf][df]);
-    }
+      for (int df = 0; df < 4; ++df) {
+        const bf16x8v bV = *(const bf16x8v*)(vt + (df * 16 + c) * FA_BN
+                                             + kk * 32 + g * 8);
+#pragma unroll
+        for (int rf = 0; rf < 2; ++rf)
+          acc[rf][df] = MFMA16(aP[rf][kk], bV, acc[rf][df]);
+      }
   }
 
   __hip_bfloat16* ob = o + bh * N * FA_D;
@@ -254,9 +258,9 @@ k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
             const float* __restrict__ lse, const float* __restrict__ delta,
             __hip_bfloat16* __restrict__ dq, int64_t N, float scale,
             int causal) {
-  // lds: Kt double buffer [2][64][32] + per-wave dS [4][32][32]
-  __shared__ __align__(16) __hip_bfloat16 lds[2 * FA_D * FA_BK
-                                              + 4 * FA_BK * FA_BK];
+  // lds: Kt double buffer [2][64][64] + per-wave dS [4][32][64]
+  __shared__ __align__(16) __hip_bfloat16 lds[2 * FA_D * FA_BN
+                                              + 4 * 32 * FA_BN];
   const int lane = threadIdx.x & 63;
   const int wv = threadIdx.x >> 6;
   const int c = lane & 15;
@@ -268,7 +272,7 @@ k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
   const __hip_bfloat16* kb = k + bh * N * FA_D;
   const __hip_bfloat16* vb = v + bh * N * FA_D;
   const __hip_bfloat16* dob = dout + bh * N * FA_D;
-  __hip_bfloat16* sbuf = lds + 2 * FA_D * FA_BK + wv * FA_BK * FA_BK;
+  __hip_bfloat16* sbuf = lds + 2 * FA_D * FA_BN + wv * 32 * FA_BN;
 
   bf16x8v aQ[2][2], aDO[2][2];
   float lse_r[2][4], dl_r[2][4];
@@ -293,30 +297,30 @@ k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
     for (int df = 0; df < 4; ++df) acc[rf][df] = (f32x4v)(0.0f);
 
   const int64_t kend = causal ? min(N, m0 + FA_BM) : N;
-  const int64_t ntiles = (kend + FA_BK - 1) / FA_BK;
+  const int64_t ntiles = (kend + FA_BN - 1) / FA_BN;
   if (ntiles <= 0) return;
   fa_fill_t(kb, 0, N, lds);
   for (int64_t t = 0; t < ntiles; ++t) {
-    const int64_t k0 = t * FA_BK;
+    const int64_t k0 = t * FA_BN;
     __syncthreads();
     if (t + 1 < ntiles)
-      fa_fill_t(kb, (t + 1) * FA_BK, N, lds + ((t + 1) & 1) * FA_D * FA_BK);
-    const __hip_bfloat16* kt = lds + (t & 1) * FA_D * FA_BK;
+      fa_fill_t(kb, (t + 1) * FA_BN, N, lds + ((t + 1) & 1) * FA_D * FA_BN);
+    const __hip_bfloat16* kt = lds + (t & 1) * FA_D * FA_BN;
     if (causal && k0 > row0 + 31) continue;
 
     // S and dP in one pass over kk
-    f32x4v S[2][2], dP[2][2];
+    f32x4v S[2][4], dP[2][4];
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
 #pragma unroll
-      for (int jf = 0; jf < 2; ++jf) {
+      for (int jf = 0; jf < 4; ++jf) {
         S[rf][jf] = (f32x4v)(0.0f);
         dP[rf][jf] = (f32x4v)(0.0f);
       }
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
 #pragma unroll
-      for (int jf = 0; jf < 2; ++jf) {
+      for (int jf = 0; jf < 4; ++jf) {
         const bf16x8v bK = fa_ldrow8(kb, k0 + jf * 16 + c, N,
                                      kk * 32 + g * 8);
         const bf16x8v bV = fa_ldrow8(vb, k0 + jf * 16 + c, N,
@@ -336,7 +340,7 @@ k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
         const int64_t qrow = row0 + rf * 16 + g * 4 + r;
         const int64_t cmax = causal ? min(N, qrow + 1) : N;
 #pragma unroll
-        for (int jf = 0; jf < 2; ++jf) {
+        for (int jf = 0; jf < 4; ++jf) {
           float s = S[rf][jf][r] * scale;
           if (k0 + jf * 16 + c >= cmax) s = FA_MASK;
           const float p = __expf(s - lse_r[rf][r]);
@@ -347,23 +351,28 @@ k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
 #pragma unroll
-      for (int jf = 0; jf < 2; ++jf)
+      for (int jf = 0; jf < 4; ++jf)
 #pragma unroll
         for (int r = 0; r < 4; ++r)
-          sbuf[(rf * 16 + g * 4 + r) * FA_BK + jf * 16 + c] =
+          sbuf[(rf * 16 + g * 4 + r) * FA_BN + jf * 16 + c] =
               (__hip_bfloat16)S[rf][jf][r];
-    bf16x8v aDS[2];
+    bf16x8v aDS[2][2];
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
-      aDS[rf] = *(const bf16x8v*)(sbuf + (rf * 16 + c) * FA_BK + g * 8);
 #pragma unroll
-    for (int df = 0; df < 4; ++df) {
-      const bf16x8v bKt =
-          *(const bf16x8v*)(kt + (df * 16 + c) * FA_BK + g * 8);
+      for (int kk = 0; kk < 2; ++kk)
+        aDS[rf][kk] = *(const bf16x8v*)(sbuf + (rf * 16 + c) * FA_BN
+                                        + kk * 32 + g * 8);
 #pragma unroll
-      for (int rf = 0; rf < 2; ++rf)
-        acc[rf][df] = MFMA16(aDS[rf], bKt, acc[rf][df]);
-    }
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int df = 0; df < 4; ++df) {
+        const bf16x8v bKt = *(const bf16x8v*)(kt + (df * 16 + c) * FA_BN
+                                              + kk * 32 + g * 8);
+#pragma unroll
+        for (int rf = 0; rf < 2; ++rf)
+          acc[rf][df] = MFMA16(aDS[rf][kk], bKt, acc[rf][df]);
+      }
   }
 
   __hip_bfloat16* dqb = dq + bh * N * FA_D;
@@ -388,9 +397,9 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
              const float* __restrict__ lse, const float* __restrict__ delta,
              __hip_bfloat16* __restrict__ dk, __hip_bfloat16* __restrict__ dv,
              int64_t N, float scale, int causal) {
-  // lds: Qt [64][32] + dOt [64][32] + per-wave P/dS [4][32][32]
-  __shared__ __align__(16) __hip_bfloat16 lds[2 * FA_D * FA_BK
-                                              + 4 * FA_BK * FA_BK];
+  // lds: Qt [64][64] + dOt [64][64] + per-wave P/dS [4][32][64]
+  __shared__ __align__(16) __hip_bfloat16 lds[2 * FA_D * FA_BN
+                                              + 4 * 32 * FA_BN];
   const int lane = threadIdx.x & 63;
   const int wv = threadIdx.x >> 6;
   const int c = lane & 15;
@@ -402,9 +411,9 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
   const __hip_bfloat16* kb = k + bh * N * FA_D;
   const __hip_bfloat16* vb = v + bh * N * FA_D;
   const __hip_bfloat16* dob = dout + bh * N * FA_D;
-  __hip_bfloat16* qt = lds;                       // [64][32]
-  __hip_bfloat16* dot = lds + FA_D * FA_BK;       // [64][32]
-  __hip_bfloat16* sbuf = lds + 2 * FA_D * FA_BK + wv * FA_BK * FA_BK;
+  __hip_bfloat16* qt = lds;                       // [64][64]
+  __hip_bfloat16* dot = lds + FA_D * FA_BN;       // [64][64]
+  __hip_bfloat16* sbuf = lds + 2 * FA_D * FA_BN + wv * 32 * FA_BN;
 
   bf16x8v aK[2][2], aV[2][2];
 #pragma unroll
@@ -424,34 +433,34 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
     }
 
   const int64_t q0_start = causal ? m0 : 0;
-  for (int64_t q0 = q0_start; q0 < N; q0 += FA_BK) {
+  for (int64_t q0 = q0_start; q0 < N; q0 += FA_BN) {
     __syncthreads();  // prior tile's Qt/dOt reads complete
     fa_fill_t(qb, q0, N, qt);
     fa_fill_t(dob, q0, N, dot);
     __syncthreads();
-    if (causal && q0 + 31 < key0) continue;  // entirely below diagonal
+    if (causal && q0 + FA_BN - 1 < key0) continue;  // below diagonal
 
-    float lse_c[2], dl_c[2];
+    float lse_c[4], dl_c[4];
 #pragma unroll
-    for (int jf = 0; jf < 2; ++jf) {
+    for (int jf = 0; jf < 4; ++jf) {
       const int64_t qcol = q0 + jf * 16 + c;
       lse_c[jf] = (qcol < N) ? lse[bh * N + qcol] : 0.0f;
       dl_c[jf] = (qcol < N) ? delta[bh * N + qcol] : 0.0f;
     }
 
     // S^T = K @ Q^T and dP^T = V @ dO^T (B-frags direct from rows)
-    f32x4v St[2][2], dPt[2][2];
+    f32x4v St[2][4], dPt[2][4];
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
 #pragma unroll
-      for (int jf = 0; jf < 2; ++jf) {
+      for (int jf = 0; jf < 4; ++jf) {
         St[rf][jf] = (f32x4v)(0.0f);
         dPt[rf][jf] = (f32x4v)(0.0f);
       }
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
 #pragma unroll
-      for (int jf = 0; jf < 2; ++jf) {
+      for (int jf = 0; jf < 4; ++jf) {
         const bf16x8v bQ = fa_ldrow8(qb, q0 + jf * 16 + c, N,
                                      kk * 32 + g * 8);
         const bf16x8v bDO = fa_ldrow8(dob, q0 + jf * 16 + c, N,
@@ -464,67 +473,68 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
       }
     }
 
-    // P^T = exp(s*scale - lse[q]);  dS^T = P^T * (dP^T - delta[q])
-    float Pt[2][2][4];
+    // P^T = exp(s*scale - lse[q]); write P^T to LDS; dS^T kept in regs
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int64_t keyrow = key0 + rf * 16 + g * 4 + r;
 #pragma unroll
-        for (int jf = 0; jf < 2; ++jf) {
+        for (int jf = 0; jf < 4; ++jf) {
           const int64_t qcol = q0 + jf * 16 + c;
           float s = St[rf][jf][r] * scale;
           if (qcol >= N || (causal && qcol < keyrow)) s = FA_MASK;
           const float p = __expf(s - lse_c[jf]);
-          Pt[rf][jf][r] = p;
+          sbuf[(rf * 16 + g * 4 + r) * FA_BN + jf * 16 + c] =
+              (__hip_bfloat16)p;
           St[rf][jf][r] = p * (dPt[rf][jf][r] - dl_c[jf]);
         }
       }
-
-    // P^T -> LDS -> A-frags; dV += P^T @ dO (B from dOt image)
+    bf16x8v aPT[2][2];
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
 #pragma unroll
-      for (int jf = 0; jf < 2; ++jf)
+      for (int kk = 0; kk < 2; ++kk)
+        aPT[rf][kk] = *(const bf16x8v*)(sbuf + (rf * 16 + c) * FA_BN
+                                        + kk * 32 + g * 8);
+    // dV += P^T @ dO (B from dOt image)
 #pragma unroll
-        for (int r = 0; r < 4; ++r)
-          sbuf[(rf * 16 + g * 4 + r) * FA_BK + jf * 16 + c] =
-              (__hip_bfloat16)Pt[rf][jf][r];
-    bf16x8v aPT[2];
+    for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
-    for (int rf = 0; rf < 2; ++rf)
-      aPT[rf] = *(const bf16x8v*)(sbuf + (rf * 16 + c) * FA_BK + g * 8);
+      for (int df = 0; df < 4; ++df) {
+        const bf16x8v bDOt = *(const bf16x8v*)(dot + (df * 16 + c) * FA_BN
+                                               + kk * 32 + g * 8);
 #pragma unroll
-    for (int df = 0; df < 4; ++df) {
-      const bf16x8v bDOt =
-          *(const bf16x8v*)(dot + (df * 16 + c) * FA_BK + g * 8);
-#pragma unroll
-      for (int rf = 0; rf < 2; ++rf)
-        dV[rf][df] = MFMA16(aPT[rf], bDOt, dV[rf][df]);
-    }
+        for (int rf = 0; rf < 2; ++rf)
+          dV[rf][df] = MFMA16(aPT[rf][kk], bDOt, dV[rf][df]);
+      }
 
     // dS^T -> LDS -> A-frags; dK += dS^T @ Q (B from Qt image)
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
 #pragma unroll
-      for (int jf = 0; jf < 2; ++jf)
+      for (int jf = 0; jf < 4; ++jf)
 #pragma unroll
         for (int r = 0; r < 4; ++r)
-          sbuf[(rf * 16 + g * 4 + r) * FA_BK + jf * 16 + c] =
+          sbuf[(rf * 16 + g * 4 + r) * FA_BN + jf * 16 + c] =
               (__hip_bfloat16)St[rf][jf][r];
-    bf16x8v aDST[2];
+    bf16x8v aDST[2][2];
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
-      aDST[rf] = *(const bf16x8v*)(sbuf + (rf * 16 + c) * FA_BK + g * 8);
 #pragma unroll
-    for (int df = 0; df < 4; ++df) {
-      const bf16x8v bQt =
-          *(const bf16x8v*)(qt + (df * 16 + c) * FA_BK + g * 8);
+      for (int kk = 0; kk < 2; ++kk)
+        aDST[rf][kk] = *(const bf16x8v*)(sbuf + (rf * 16 + c) * FA_BN
+                                         + kk * 32 + g * 8);
 #pragma unroll
-      for (int rf = 0; rf < 2; ++rf)
-        dK[rf][df] = MFMA16(aDST[rf], bQt, dK[rf][df]);
-    }
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int df = 0; df < 4; ++df) {
+        const bf16x8v bQt = *(const bf16x8v*)(qt + (df * 16 + c) * FA_BN
+                                              + kk * 32 + g * 8);
+#pragma unroll
+        for (int rf = 0; rf < 2; ++rf)
+          dK[rf][df] = MFMA16(aDST[rf][kk], bQt, dK[rf][df]);
+      }
   }
 
   __hip_bfloat16* dkb = dk + bh * N * FA_D;
@@ -546,8 +556,8 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
 
 // ---------------------------------------------------------------------------
 // fragment-layout self-check: one wave computes C[16][16] = A[16][32] @
-// B^T? no — B given as [32][16] column-slices per the assumed B layout.
-// Host compares against a reference matmul; a layout mismatch fails loudly.
+// B[32][16] via one mfma using the assumed lane maps.  Host compares against
+// a reference matmul; a layout mismatch fails loudly.
 // ---------------------------------------------------------------------------
 
 __global__ void k_fa_selfcheck(const __hip_bfloat16* __restrict__ A,
@@ -560,8 +570,8 @@ __global__ void k_fa_selfcheck(const __hip_bfloat16* __restrict__ A,
   bf16x8v a, b;
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
-    a[j] = A[c * 32 + g * 8 + j];      // A[row=c][k]
-    b[j] = B[(g * 8 + j) * 16 + c];    // B[k][col=c]
+    a[j] = (__bf16)(float)A[c * 32 + g * 8 + j];   // A[row=c][k]
+    b[j] = (__bf16)(float)B[(g * 8 + j) * 16 + c]; // B[k][col=c]
   }
   f32x4v acc = (f32x4v)(0.0f);
   acc = MFMA16(a, b, acc);
