@@ -40,7 +40,11 @@ typedef __attribute__((ext_vector_type(4))) float f32x4v;
 
 #define FA_D 64
 #define FA_BM 128   // rows (fwd/dq) or keys (dkv) per 4-wave block
-#define FA_BN 64    // kv (fwd/dq) or q (dkv) tile width
+#define FA_BN 64    // fwd kv tile width
+#define FA_BNB 32   // bwd tile width: the bwd kernels carry 2-3x the live
+                    // state of fwd (two C-tiles + two accumulators), and at
+                    // BN=64 their VGPR+AGPR total passes 256 -> 1 wave/SIMD
+                    // (measured regression); 32 keeps them at 2 waves/SIMD
 #define FA_MASK -1e30f
 
 __device__ __forceinline__ bf16x8v fa_zero8() {
@@ -72,6 +76,18 @@ __device__ __forceinline__ void fa_fill_t(const __hip_bfloat16* src,
     for (int j = 0; j < 8; ++j)
       dstT[(d0 + h * 8 + j) * FA_BN + k] = (__hip_bfloat16)(float)v[j];
   }
+}
+
+// 32-row variant for the bwd kernels: rows [r0, r0+32) -> dstT[64][32]
+__device__ __forceinline__ void fa_fill_t32(const __hip_bfloat16* src,
+                                            int64_t r0, int64_t nrows,
+                                            __hip_bfloat16* dstT) {
+  const int k = threadIdx.x >> 3;        // 0..31 source row in tile
+  const int d0 = (threadIdx.x & 7) * 8;  // 0..56
+  const bf16x8v v = fa_ldrow8(src, r0 + k, nrows, d0);
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    dstT[(d0 + j) * FA_BNB + k] = (__hip_bfloat16)(float)v[j];
 }
 
 // ---------------------------------------------------------------------------
@@ -132,23 +148,27 @@ k_fa_fwd(const __hip_bfloat16* __restrict__ q,
     const __hip_bfloat16* vt = lds + (t & 1) * FA_D * FA_BN;
     if (causal && k0 > row0 + 31) continue;  // above this wave's diagonal
 
-    // S = Q @ K^T  (B-frag: 8 consecutive d of key row -> direct load)
+    // S = Q @ K^T  (B-frag: 8 consecutive d of key row -> direct load).
+    // Load ALL 8 K fragments first so their L2 latencies overlap (a load
+    // adjacent to its consuming MFMA costs one serialized round trip each).
+    bf16x8v bK[4][2];
+#pragma unroll
+    for (int jf = 0; jf < 4; ++jf)
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+        bK[jf][kk] = fa_ldrow8(kb, k0 + jf * 16 + c, N, kk * 32 + g * 8);
     f32x4v S[2][4];
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
 #pragma unroll
       for (int jf = 0; jf < 4; ++jf) S[rf][jf] = (f32x4v)(0.0f);
 #pragma unroll
-    for (int kk = 0; kk < 2; ++kk) {
+    for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
-      for (int jf = 0; jf < 4; ++jf) {
-        const bf16x8v bK = fa_ldrow8(kb, k0 + jf * 16 + c, N,
-                                     kk * 32 + g * 8);
+      for (int jf = 0; jf < 4; ++jf)
 #pragma unroll
         for (int rf = 0; rf < 2; ++rf)
-          S[rf][jf] = MFMA16(aQ[rf][kk], bK, S[rf][jf]);
-      }
-    }
+          S[rf][jf] = MFMA16(aQ[rf][kk], bK[jf][kk], S[rf][jf]);
 
     // scale + mask + online softmax
 #pragma unroll
@@ -258,9 +278,9 @@ k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
             const float* __restrict__ lse, const float* __restrict__ delta,
             __hip_bfloat16* __restrict__ dq, int64_t N, float scale,
             int causal) {
-  // lds: Kt double buffer [2][64][64] + per-wave dS [4][32][64]
-  __shared__ __align__(16) __hip_bfloat16 lds[2 * FA_D * FA_BN
-                                              + 4 * 32 * FA_BN];
+  // lds: Kt double buffer [2][64][32] + per-wave dS [4][32][32]
+  __shared__ __align__(16) __hip_bfloat16 lds[2 * FA_D * FA_BNB
+                                              + 4 * 32 * FA_BNB];
   const int lane = threadIdx.x & 63;
   const int wv = threadIdx.x >> 6;
   const int c = lane & 15;
@@ -272,7 +292,7 @@ k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
   const __hip_bfloat16* kb = k + bh * N * FA_D;
   const __hip_bfloat16* vb = v + bh * N * FA_D;
   const __hip_bfloat16* dob = dout + bh * N * FA_D;
-  __hip_bfloat16* sbuf = lds + 2 * FA_D * FA_BN + wv * 32 * FA_BN;
+  __hip_bfloat16* sbuf = lds + 2 * FA_D * FA_BNB + wv * 32 * FA_BNB;
 
   bf16x8v aQ[2][2], aDO[2][2];
   float lse_r[2][4], dl_r[2][4];
@@ -297,41 +317,44 @@ k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
     for (int df = 0; df < 4; ++df) acc[rf][df] = (f32x4v)(0.0f);
 
   const int64_t kend = causal ? min(N, m0 + FA_BM) : N;
-  const int64_t ntiles = (kend + FA_BN - 1) / FA_BN;
+  const int64_t ntiles = (kend + FA_BNB - 1) / FA_BNB;
   if (ntiles <= 0) return;
-  fa_fill_t(kb, 0, N, lds);
+  fa_fill_t32(kb, 0, N, lds);
   for (int64_t t = 0; t < ntiles; ++t) {
-    const int64_t k0 = t * FA_BN;
+    const int64_t k0 = t * FA_BNB;
     __syncthreads();
     if (t + 1 < ntiles)
-      fa_fill_t(kb, (t + 1) * FA_BN, N, lds + ((t + 1) & 1) * FA_D * FA_BN);
-    const __hip_bfloat16* kt = lds + (t & 1) * FA_D * FA_BN;
+      fa_fill_t32(kb, (t + 1) * FA_BNB,
+                  N, lds + ((t + 1) & 1) * FA_D * FA_BNB);
+    const __hip_bfloat16* kt = lds + (t & 1) * FA_D * FA_BNB;
     if (causal && k0 > row0 + 31) continue;
 
-    // S and dP in one pass over kk
-    f32x4v S[2][4], dP[2][4];
+    // S and dP in one pass over kk (batched B-frag loads: latencies overlap)
+    bf16x8v bK[2][2], bV[2][2];
+#pragma unroll
+    for (int jf = 0; jf < 2; ++jf)
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        bK[jf][kk] = fa_ldrow8(kb, k0 + jf * 16 + c, N, kk * 32 + g * 8);
+        bV[jf][kk] = fa_ldrow8(vb, k0 + jf * 16 + c, N, kk * 32 + g * 8);
+      }
+    f32x4v S[2][2], dP[2][2];
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
 #pragma unroll
-      for (int jf = 0; jf < 4; ++jf) {
+      for (int jf = 0; jf < 2; ++jf) {
         S[rf][jf] = (f32x4v)(0.0f);
         dP[rf][jf] = (f32x4v)(0.0f);
       }
 #pragma unroll
-    for (int kk = 0; kk < 2; ++kk) {
+    for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
-      for (int jf = 0; jf < 4; ++jf) {
-        const bf16x8v bK = fa_ldrow8(kb, k0 + jf * 16 + c, N,
-                                     kk * 32 + g * 8);
-        const bf16x8v bV = fa_ldrow8(vb, k0 + jf * 16 + c, N,
-                                     kk * 32 + g * 8);
+      for (int jf = 0; jf < 2; ++jf)
 #pragma unroll
         for (int rf = 0; rf < 2; ++rf) {
-          S[rf][jf] = MFMA16(aQ[rf][kk], bK, S[rf][jf]);
-          dP[rf][jf] = MFMA16(aDO[rf][kk], bV, dP[rf][jf]);
+          S[rf][jf] = MFMA16(aQ[rf][kk], bK[jf][kk], S[rf][jf]);
+          dP[rf][jf] = MFMA16(aDO[rf][kk], bV[jf][kk], dP[rf][jf]);
         }
-      }
-    }
     // dS = P * (dP - delta)
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
@@ -340,7 +363,7 @@ k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
         const int64_t qrow = row0 + rf * 16 + g * 4 + r;
         const int64_t cmax = causal ? min(N, qrow + 1) : N;
 #pragma unroll
-        for (int jf = 0; jf < 4; ++jf) {
+        for (int jf = 0; jf < 2; ++jf) {
           float s = S[rf][jf][r] * scale;
           if (k0 + jf * 16 + c >= cmax) s = FA_MASK;
           const float p = __expf(s - lse_r[rf][r]);
@@ -351,28 +374,23 @@ k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
 #pragma unroll
-      for (int jf = 0; jf < 4; ++jf)
+      for (int jf = 0; jf < 2; ++jf)
 #pragma unroll
         for (int r = 0; r < 4; ++r)
-          sbuf[(rf * 16 + g * 4 + r) * FA_BN + jf * 16 + c] =
+          sbuf[(rf * 16 + g * 4 + r) * FA_BNB + jf * 16 + c] =
               (__hip_bfloat16)S[rf][jf][r];
-    bf16x8v aDS[2][2];
+    bf16x8v aDS[2];
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
+      aDS[rf] = *(const bf16x8v*)(sbuf + (rf * 16 + c) * FA_BNB + g * 8);
 #pragma unroll
-      for (int kk = 0; kk < 2; ++kk)
-        aDS[rf][kk] = *(const bf16x8v*)(sbuf + (rf * 16 + c) * FA_BN
-                                        + kk * 32 + g * 8);
+    for (int df = 0; df < 4; ++df) {
+      const bf16x8v bKt =
+          *(const bf16x8v*)(kt + (df * 16 + c) * FA_BNB + g * 8);
 #pragma unroll
-    for (int kk = 0; kk < 2; ++kk)
-#pragma unroll
-      for (int df = 0; df < 4; ++df) {
-        const bf16x8v bKt = *(const bf16x8v*)(kt + (df * 16 + c) * FA_BN
-                                              + kk * 32 + g * 8);
-#pragma unroll
-        for (int rf = 0; rf < 2; ++rf)
-          acc[rf][df] = MFMA16(aDS[rf][kk], bKt, acc[rf][df]);
-      }
+      for (int rf = 0; rf < 2; ++rf)
+        acc[rf][df] = MFMA16(aDS[rf], bKt, acc[rf][df]);
+    }
   }
 
   __hip_bfloat16* dqb = dq + bh * N * FA_D;
@@ -397,9 +415,9 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
              const float* __restrict__ lse, const float* __restrict__ delta,
              __hip_bfloat16* __restrict__ dk, __hip_bfloat16* __restrict__ dv,
              int64_t N, float scale, int causal) {
-  // lds: Qt [64][64] + dOt [64][64] + per-wave P/dS [4][32][64]
-  __shared__ __align__(16) __hip_bfloat16 lds[2 * FA_D * FA_BN
-                                              + 4 * 32 * FA_BN];
+  // lds: Qt [64][32] + dOt [64][32] + per-wave P/dS [4][32][32]
+  __shared__ __align__(16) __hip_bfloat16 lds[2 * FA_D * FA_BNB
+                                              + 4 * 32 * FA_BNB];
   const int lane = threadIdx.x & 63;
   const int wv = threadIdx.x >> 6;
   const int c = lane & 15;
@@ -411,9 +429,9 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
   const __hip_bfloat16* kb = k + bh * N * FA_D;
   const __hip_bfloat16* vb = v + bh * N * FA_D;
   const __hip_bfloat16* dob = dout + bh * N * FA_D;
-  __hip_bfloat16* qt = lds;                       // [64][64]
-  __hip_bfloat16* dot = lds + FA_D * FA_BN;       // [64][64]
-  __hip_bfloat16* sbuf = lds + 2 * FA_D * FA_BN + wv * 32 * FA_BN;
+  __hip_bfloat16* qt = lds;                        // [64][32]
+  __hip_bfloat16* dot = lds + FA_D * FA_BNB;       // [64][32]
+  __hip_bfloat16* sbuf = lds + 2 * FA_D * FA_BNB + wv * 32 * FA_BNB;
 
   bf16x8v aK[2][2], aV[2][2];
 #pragma unroll
@@ -433,45 +451,47 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
     }
 
   const int64_t q0_start = causal ? m0 : 0;
-  for (int64_t q0 = q0_start; q0 < N; q0 += FA_BN) {
+  for (int64_t q0 = q0_start; q0 < N; q0 += FA_BNB) {
     __syncthreads();  // prior tile's Qt/dOt reads complete
-    fa_fill_t(qb, q0, N, qt);
-    fa_fill_t(dob, q0, N, dot);
+    fa_fill_t32(qb, q0, N, qt);
+    fa_fill_t32(dob, q0, N, dot);
     __syncthreads();
-    if (causal && q0 + FA_BN - 1 < key0) continue;  // below diagonal
+    if (causal && q0 + FA_BNB - 1 < key0) continue;  // below diagonal
 
-    float lse_c[4], dl_c[4];
+    float lse_c[2], dl_c[2];
 #pragma unroll
-    for (int jf = 0; jf < 4; ++jf) {
+    for (int jf = 0; jf < 2; ++jf) {
       const int64_t qcol = q0 + jf * 16 + c;
       lse_c[jf] = (qcol < N) ? lse[bh * N + qcol] : 0.0f;
       dl_c[jf] = (qcol < N) ? delta[bh * N + qcol] : 0.0f;
     }
 
-    // S^T = K @ Q^T and dP^T = V @ dO^T (B-frags direct from rows)
-    f32x4v St[2][4], dPt[2][4];
+    // S^T = K @ Q^T and dP^T = V @ dO^T (batched direct B-frag loads)
+    bf16x8v bQ[2][2], bDO[2][2];
+#pragma unroll
+    for (int jf = 0; jf < 2; ++jf)
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        bQ[jf][kk] = fa_ldrow8(qb, q0 + jf * 16 + c, N, kk * 32 + g * 8);
+        bDO[jf][kk] = fa_ldrow8(dob, q0 + jf * 16 + c, N, kk * 32 + g * 8);
+      }
+    f32x4v St[2][2], dPt[2][2];
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
 #pragma unroll
-      for (int jf = 0; jf < 4; ++jf) {
+      for (int jf = 0; jf < 2; ++jf) {
         St[rf][jf] = (f32x4v)(0.0f);
         dPt[rf][jf] = (f32x4v)(0.0f);
       }
 #pragma unroll
-    for (int kk = 0; kk < 2; ++kk) {
+    for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
-      for (int jf = 0; jf < 4; ++jf) {
-        const bf16x8v bQ = fa_ldrow8(qb, q0 + jf * 16 + c, N,
-                                     kk * 32 + g * 8);
-        const bf16x8v bDO = fa_ldrow8(dob, q0 + jf * 16 + c, N,
-                                      kk * 32 + g * 8);
+      for (int jf = 0; jf < 2; ++jf)
 #pragma unroll
         for (int rf = 0; rf < 2; ++rf) {
-          St[rf][jf] = MFMA16(aK[rf][kk], bQ, St[rf][jf]);
-          dPt[rf][jf] = MFMA16(aV[rf][kk], bDO, dPt[rf][jf]);
+          St[rf][jf] = MFMA16(aK[rf][kk], bQ[jf][kk], St[rf][jf]);
+          dPt[rf][jf] = MFMA16(aV[rf][kk], bDO[jf][kk], dPt[rf][jf]);
         }
-      }
-    }
 
     // P^T = exp(s*scale - lse[q]); write P^T to LDS; dS^T kept in regs
 #pragma unroll
@@ -480,61 +500,51 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
       for (int r = 0; r < 4; ++r) {
         const int64_t keyrow = key0 + rf * 16 + g * 4 + r;
 #pragma unroll
-        for (int jf = 0; jf < 4; ++jf) {
+        for (int jf = 0; jf < 2; ++jf) {
           const int64_t qcol = q0 + jf * 16 + c;
           float s = St[rf][jf][r] * scale;
           if (qcol >= N || (causal && qcol < keyrow)) s = FA_MASK;
           const float p = __expf(s - lse_c[jf]);
-          sbuf[(rf * 16 + g * 4 + r) * FA_BN + jf * 16 + c] =
+          sbuf[(rf * 16 + g * 4 + r) * FA_BNB + jf * 16 + c] =
               (__hip_bfloat16)p;
           St[rf][jf][r] = p * (dPt[rf][jf][r] - dl_c[jf]);
         }
       }
-    bf16x8v aPT[2][2];
+    bf16x8v aPT[2];
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
-#pragma unroll
-      for (int kk = 0; kk < 2; ++kk)
-        aPT[rf][kk] = *(const bf16x8v*)(sbuf + (rf * 16 + c) * FA_BN
-                                        + kk * 32 + g * 8);
+      aPT[rf] = *(const bf16x8v*)(sbuf + (rf * 16 + c) * FA_BNB + g * 8);
     // dV += P^T @ dO (B from dOt image)
 #pragma unroll
-    for (int kk = 0; kk < 2; ++kk)
+    for (int df = 0; df < 4; ++df) {
+      const bf16x8v bDOt =
+          *(const bf16x8v*)(dot + (df * 16 + c) * FA_BNB + g * 8);
 #pragma unroll
-      for (int df = 0; df < 4; ++df) {
-        const bf16x8v bDOt = *(const bf16x8v*)(dot + (df * 16 + c) * FA_BN
-                                               + kk * 32 + g * 8);
-#pragma unroll
-        for (int rf = 0; rf < 2; ++rf)
-          dV[rf][df] = MFMA16(aPT[rf][kk], bDOt, dV[rf][df]);
-      }
+      for (int rf = 0; rf < 2; ++rf)
+        dV[rf][df] = MFMA16(aPT[rf], bDOt, dV[rf][df]);
+    }
 
     // dS^T -> LDS -> A-frags; dK += dS^T @ Q (B from Qt image)
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
 #pragma unroll
-      for (int jf = 0; jf < 4; ++jf)
+      for (int jf = 0; jf < 2; ++jf)
 #pragma unroll
         for (int r = 0; r < 4; ++r)
-          sbuf[(rf * 16 + g * 4 + r) * FA_BN + jf * 16 + c] =
+          sbuf[(rf * 16 + g * 4 + r) * FA_BNB + jf * 16 + c] =
               (__hip_bfloat16)St[rf][jf][r];
-    bf16x8v aDST[2][2];
+    bf16x8v aDST[2];
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
+      aDST[rf] = *(const bf16x8v*)(sbuf + (rf * 16 + c) * FA_BNB + g * 8);
 #pragma unroll
-      for (int kk = 0; kk < 2; ++kk)
-        aDST[rf][kk] = *(const bf16x8v*)(sbuf + (rf * 16 + c) * FA_BN
-                                         + kk * 32 + g * 8);
+    for (int df = 0; df < 4; ++df) {
+      const bf16x8v bQt =
+          *(const bf16x8v*)(qt + (df * 16 + c) * FA_BNB + g * 8);
 #pragma unroll
-    for (int kk = 0; kk < 2; ++kk)
-#pragma unroll
-      for (int df = 0; df < 4; ++df) {
-        const bf16x8v bQt = *(const bf16x8v*)(qt + (df * 16 + c) * FA_BN
-                                              + kk * 32 + g * 8);
-#pragma unroll
-        for (int rf = 0; rf < 2; ++rf)
-          dK[rf][df] = MFMA16(aDST[rf][kk], bQt, dK[rf][df]);
-      }
+      for (int rf = 0; rf < 2; ++rf)
+        dK[rf][df] = MFMA16(aDST[rf], bQt, dK[rf][df]);
+    }
   }
 
   __hip_bfloat16* dkb = dk + bh * N * FA_D;
