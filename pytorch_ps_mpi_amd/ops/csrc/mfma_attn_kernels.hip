@@ -54,6 +54,20 @@ __device__ __forceinline__ bf16x8v fa_zero8() {
   return z;
 }
 
+// XOR-swizzled LDS indexing (guide T2): every image is [row][col] bf16 with
+// an 8-element (16B) granule; granule index is XORed with the row so lanes
+// reading one column-range of many rows spread over banks.  Writes use
+// fa_swe (element), 16B vector reads use fa_swg (granule base).  mask =
+// cols/8 - 1.  Without this the scalar transpose-fill and P-roundtrip
+// writes are 4-8-way bank conflicted (measured 10-13% of wave cycles).
+__device__ __forceinline__ int fa_swe(int row, int col, int stride,
+                                      int mask) {
+  return row * stride + ((((col >> 3) ^ row) & mask) << 3) + (col & 7);
+}
+__device__ __forceinline__ int fa_swg(int row, int g8, int stride, int mask) {
+  return row * stride + (((g8 ^ row) & mask) << 3);
+}
+
 // 8 consecutive bf16 of row `row` at element offset `off` (row-guarded).
 __device__ __forceinline__ bf16x8v fa_ldrow8(const __hip_bfloat16* base,
                                              int64_t row, int64_t nrows,
@@ -62,8 +76,8 @@ __device__ __forceinline__ bf16x8v fa_ldrow8(const __hip_bfloat16* base,
   return fa_zero8();
 }
 
-// cooperative transpose fill: src rows [r0, r0+64) x 64 cols -> dstT[64][64]
-// (dstT row stride FA_BN).  256 threads, 16 elems each.
+// cooperative transpose fill: src rows [r0, r0+64) x 64 cols -> swizzled
+// dstT[64][64] (row stride FA_BN, mask 7).  256 threads, 16 elems each.
 __device__ __forceinline__ void fa_fill_t(const __hip_bfloat16* src,
                                           int64_t r0, int64_t nrows,
                                           __hip_bfloat16* dstT) {
@@ -74,11 +88,12 @@ __device__ __forceinline__ void fa_fill_t(const __hip_bfloat16* src,
     const bf16x8v v = fa_ldrow8(src, r0 + k, nrows, d0 + h * 8);
 #pragma unroll
     for (int j = 0; j < 8; ++j)
-      dstT[(d0 + h * 8 + j) * FA_BN + k] = (__hip_bfloat16)(float)v[j];
+      dstT[fa_swe(d0 + h * 8 + j, k, FA_BN, 7)] = (__hip_bfloat16)(float)v[j];
   }
 }
 
-// 32-row variant for the bwd kernels: rows [r0, r0+32) -> dstT[64][32]
+// 32-row variant for the bwd kernels: rows [r0, r0+32) -> swizzled
+// dstT[64][32] (row stride FA_BNB, mask 3)
 __device__ __forceinline__ void fa_fill_t32(const __hip_bfloat16* src,
                                             int64_t r0, int64_t nrows,
                                             __hip_bfloat16* dstT) {
@@ -87,7 +102,7 @@ __device__ __forceinline__ void fa_fill_t32(const __hip_bfloat16* src,
   const bf16x8v v = fa_ldrow8(src, r0 + k, nrows, d0);
 #pragma unroll
   for (int j = 0; j < 8; ++j)
-    dstT[(d0 + j) * FA_BNB + k] = (__hip_bfloat16)(float)v[j];
+    dstT[fa_swe(d0 + j, k, FA_BNB, 3)] = (__hip_bfloat16)(float)v[j];
 }
 
 // ---------------------------------------------------------------------------
@@ -212,23 +227,23 @@ k_fa_fwd(const __hip_bfloat16* __restrict__ q,
       for (int jf = 0; jf < 4; ++jf)
 #pragma unroll
         for (int r = 0; r < 4; ++r)
-          pbuf[(rf * 16 + g * 4 + r) * FA_BN + jf * 16 + c] =
+          pbuf[fa_swe(rf * 16 + g * 4 + r, jf * 16 + c, FA_BN, 7)] =
               (__hip_bfloat16)S[rf][jf][r];
     bf16x8v aP[2][2];
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
 #pragma unroll
       for (int kk = 0; kk < 2; ++kk)
-        aP[rf][kk] = *(const bf16x8v*)(pbuf + (rf * 16 + c) * FA_BN
-                                       + kk * 32 + g * 8);
+        aP[rf][kk] = *(const bf16x8v*)(pbuf
+            + fa_swg(rf * 16 + c, kk * 4 + g, FA_BN, 7));
 
     // O += P @ V   (B-frag from transposed V image)
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
       for (int df = 0; df < 4; ++df) {
-        const bf16x8v bV = *(const bf16x8v*)(vt + (df * 16 + c) * FA_BN
-                                             + kk * 32 + g * 8);
+        const bf16x8v bV = *(const bf16x8v*)(vt
+            + fa_swg(df * 16 + c, kk * 4 + g, FA_BN, 7));
 #pragma unroll
         for (int rf = 0; rf < 2; ++rf)
           acc[rf][df] = MFMA16(aP[rf][kk], bV, acc[rf][df]);
@@ -259,14 +274,26 @@ __global__ void __launch_bounds__(256)
 k_fa_delta(const __hip_bfloat16* __restrict__ dout,
            const __hip_bfloat16* __restrict__ o, float* __restrict__ delta,
            int64_t rows) {
+  // 8 rows per wave: lane owns 8 consecutive elements (one 16B load per
+  // tensor) of row lane>>3; dot reduced across the 8-lane row group.
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int64_t rstride = (int64_t)gridDim.x * 4;
-  for (int64_t r = (int64_t)blockIdx.x * 4 + wave; r < rows; r += rstride) {
-    float d = (float)dout[r * FA_D + lane] * (float)o[r * FA_D + lane];
+  const int rsub = lane >> 3;        // 0..7 row within the wave's 8
+  const int off = (lane & 7) * 8;
+  const int64_t rstride = (int64_t)gridDim.x * 32;
+  for (int64_t r0 = ((int64_t)blockIdx.x * 4 + wave) * 8; r0 < rows;
+       r0 += rstride) {
+    const int64_t r = r0 + rsub;
+    float d = 0.0f;
+    if (r < rows) {
+      const bf16x8v a = *(const bf16x8v*)(dout + r * FA_D + off);
+      const bf16x8v b = *(const bf16x8v*)(o + r * FA_D + off);
 #pragma unroll
-    for (int s = 1; s < 64; s <<= 1) d += __shfl_xor(d, s, 64);
-    if (lane == 0) delta[r] = d;
+      for (int j = 0; j < 8; ++j) d += (float)a[j] * (float)b[j];
+    }
+#pragma unroll
+    for (int s = 1; s < 8; s <<= 1) d += __shfl_xor(d, s, 8);
+    if ((lane & 7) == 0 && r < rows) delta[r] = d;
   }
 }
 
@@ -377,16 +404,16 @@ k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
       for (int jf = 0; jf < 2; ++jf)
 #pragma unroll
         for (int r = 0; r < 4; ++r)
-          sbuf[(rf * 16 + g * 4 + r) * FA_BNB + jf * 16 + c] =
+          sbuf[fa_swe(rf * 16 + g * 4 + r, jf * 16 + c, FA_BNB, 3)] =
               (__hip_bfloat16)S[rf][jf][r];
     bf16x8v aDS[2];
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
-      aDS[rf] = *(const bf16x8v*)(sbuf + (rf * 16 + c) * FA_BNB + g * 8);
+      aDS[rf] = *(const bf16x8v*)(sbuf + fa_swg(rf * 16 + c, g, FA_BNB, 3));
 #pragma unroll
     for (int df = 0; df < 4; ++df) {
       const bf16x8v bKt =
-          *(const bf16x8v*)(kt + (df * 16 + c) * FA_BNB + g * 8);
+          *(const bf16x8v*)(kt + fa_swg(df * 16 + c, g, FA_BNB, 3));
 #pragma unroll
       for (int rf = 0; rf < 2; ++rf)
         acc[rf][df] = MFMA16(aDS[rf], bKt, acc[rf][df]);
@@ -415,8 +442,8 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
              const float* __restrict__ lse, const float* __restrict__ delta,
              __hip_bfloat16* __restrict__ dk, __hip_bfloat16* __restrict__ dv,
              int64_t N, float scale, int causal) {
-  // lds: Qt [64][32] + dOt [64][32] + per-wave P/dS [4][32][32]
-  __shared__ __align__(16) __hip_bfloat16 lds[2 * FA_D * FA_BNB
+  // lds: double-buffered {Qt,dOt} [2][2][64][32] + per-wave P/dS [4][32][32]
+  __shared__ __align__(16) __hip_bfloat16 lds[4 * FA_D * FA_BNB
                                               + 4 * 32 * FA_BNB];
   const int lane = threadIdx.x & 63;
   const int wv = threadIdx.x >> 6;
@@ -429,9 +456,8 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
   const __hip_bfloat16* kb = k + bh * N * FA_D;
   const __hip_bfloat16* vb = v + bh * N * FA_D;
   const __hip_bfloat16* dob = dout + bh * N * FA_D;
-  __hip_bfloat16* qt = lds;                        // [64][32]
-  __hip_bfloat16* dot = lds + FA_D * FA_BNB;       // [64][32]
-  __hip_bfloat16* sbuf = lds + 2 * FA_D * FA_BNB + wv * 32 * FA_BNB;
+  // buffer b: qt at lds + b*2*IMG, dot right after (IMG = 64*32)
+  __hip_bfloat16* sbuf = lds + 4 * FA_D * FA_BNB + wv * 32 * FA_BNB;
 
   bf16x8v aK[2][2], aV[2][2];
 #pragma unroll
@@ -451,11 +477,20 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
     }
 
   const int64_t q0_start = causal ? m0 : 0;
-  for (int64_t q0 = q0_start; q0 < N; q0 += FA_BNB) {
-    __syncthreads();  // prior tile's Qt/dOt reads complete
-    fa_fill_t32(qb, q0, N, qt);
-    fa_fill_t32(dob, q0, N, dot);
-    __syncthreads();
+  const int64_t ntiles = (N - q0_start + FA_BNB - 1) / FA_BNB;
+  if (ntiles <= 0) return;
+  fa_fill_t32(qb, q0_start, N, lds);
+  fa_fill_t32(dob, q0_start, N, lds + FA_D * FA_BNB);
+  for (int64_t t = 0; t < ntiles; ++t) {
+    const int64_t q0 = q0_start + t * FA_BNB;
+    __syncthreads();  // buffer t&1 filled; prior tile's reads complete
+    if (t + 1 < ntiles) {
+      __hip_bfloat16* nb = lds + ((t + 1) & 1) * 2 * FA_D * FA_BNB;
+      fa_fill_t32(qb, q0 + FA_BNB, N, nb);
+      fa_fill_t32(dob, q0 + FA_BNB, N, nb + FA_D * FA_BNB);
+    }
+    const __hip_bfloat16* qt = lds + (t & 1) * 2 * FA_D * FA_BNB;
+    const __hip_bfloat16* dot = qt + FA_D * FA_BNB;
     if (causal && q0 + FA_BNB - 1 < key0) continue;  // below diagonal
 
     float lse_c[2], dl_c[2];
@@ -505,7 +540,7 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
           float s = St[rf][jf][r] * scale;
           if (qcol >= N || (causal && qcol < keyrow)) s = FA_MASK;
           const float p = __expf(s - lse_c[jf]);
-          sbuf[(rf * 16 + g * 4 + r) * FA_BNB + jf * 16 + c] =
+          sbuf[fa_swe(rf * 16 + g * 4 + r, jf * 16 + c, FA_BNB, 3)] =
               (__hip_bfloat16)p;
           St[rf][jf][r] = p * (dPt[rf][jf][r] - dl_c[jf]);
         }
@@ -513,12 +548,12 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
     bf16x8v aPT[2];
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
-      aPT[rf] = *(const bf16x8v*)(sbuf + (rf * 16 + c) * FA_BNB + g * 8);
+      aPT[rf] = *(const bf16x8v*)(sbuf + fa_swg(rf * 16 + c, g, FA_BNB, 3));
     // dV += P^T @ dO (B from dOt image)
 #pragma unroll
     for (int df = 0; df < 4; ++df) {
       const bf16x8v bDOt =
-          *(const bf16x8v*)(dot + (df * 16 + c) * FA_BNB + g * 8);
+          *(const bf16x8v*)(dot + fa_swg(df * 16 + c, g, FA_BNB, 3));
 #pragma unroll
       for (int rf = 0; rf < 2; ++rf)
         dV[rf][df] = MFMA16(aPT[rf], bDOt, dV[rf][df]);
@@ -531,16 +566,16 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
       for (int jf = 0; jf < 2; ++jf)
 #pragma unroll
         for (int r = 0; r < 4; ++r)
-          sbuf[(rf * 16 + g * 4 + r) * FA_BNB + jf * 16 + c] =
+          sbuf[fa_swe(rf * 16 + g * 4 + r, jf * 16 + c, FA_BNB, 3)] =
               (__hip_bfloat16)St[rf][jf][r];
     bf16x8v aDST[2];
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
-      aDST[rf] = *(const bf16x8v*)(sbuf + (rf * 16 + c) * FA_BNB + g * 8);
+      aDST[rf] = *(const bf16x8v*)(sbuf + fa_swg(rf * 16 + c, g, FA_BNB, 3));
 #pragma unroll
     for (int df = 0; df < 4; ++df) {
       const bf16x8v bQt =
-          *(const bf16x8v*)(qt + (df * 16 + c) * FA_BNB + g * 8);
+          *(const bf16x8v*)(qt + fa_swg(df * 16 + c, g, FA_BNB, 3));
 #pragma unroll
       for (int rf = 0; rf < 2; ++rf)
         dK[rf][df] = MFMA16(aDST[rf], bQt, dK[rf][df]);
