@@ -69,7 +69,13 @@ def fused_sdpa(q, k, v, is_causal=False):
             and q.dtype == torch.bfloat16
             and q.dim() == 4 and q.shape[-1] == 64
             and q.shape == k.shape == v.shape):
+        mfma = impl == "mfma"
+        if mfma and (q.stride(-1) == 1 and q.stride(0) % 8 == 0
+                     and q.stride(1) % 8 == 0 and q.stride(2) % 8 == 0
+                     and q.stride() == k.stride() == v.stride()):
+            # the MFMA kernels read strided q/k/v natively (e.g. head
+            # slices of a fused qkv projection) — no copies
+            return _FusedSDPA.apply(q, k, v, bool(is_causal), True)
         return _FusedSDPA.apply(q.contiguous(), k.contiguous(),
-                                v.contiguous(), bool(is_causal),
-                                impl == "mfma")
+                                v.contiguous(), bool(is_causal), mfma)
     return F.scaled_dot_product_attention(q, k, v, is_causal=is_causal)

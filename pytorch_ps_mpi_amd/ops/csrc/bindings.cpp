@@ -77,11 +77,12 @@ int ps_attn_bwd(void* stream, const void* q, const void* k, const void* v,
                 int64_t N, float scale, int causal);
 int ps_fa_fwd(void* stream, const void* q, const void* k, const void* v,
               void* o, float* lse, int64_t BH, int64_t N, float scale,
-              int causal);
+              int causal, int64_t H, int64_t sB, int64_t sH, int64_t sN);
 int ps_fa_bwd(void* stream, const void* q, const void* k, const void* v,
               const void* o, const void* dout, const float* lse, float* delta,
               void* dq, void* dk, void* dv, int64_t BH, int64_t N,
-              float scale, int causal);
+              float scale, int causal, int64_t H, int64_t sB, int64_t sH,
+              int64_t sN);
 int ps_fa_selfcheck(void* stream, const void* a, const void* b, float* c);
 }
 
@@ -475,15 +476,31 @@ void attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o,
            "attn_bwd");
 }
 
+// q,k,v: [B,H,N,64] bf16 with IDENTICAL strides, unit d-stride and 16B-
+// aligned rows (covers contiguous tensors and head-slices of a fused qkv
+// projection, so the model path never copies); o/dq/dk/dv are contiguous.
+static void fa_check_qkv(const at::Tensor& q, const at::Tensor& k,
+                         const at::Tensor& v) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16 &&
+                  q.dim() == 4 && q.size(3) == 64,
+              "q must be [B,H,N,64] bf16");
+  TORCH_CHECK(q.stride(3) == 1 && q.stride(2) % 8 == 0 &&
+                  q.stride(1) % 8 == 0 && q.stride(0) % 8 == 0,
+              "q rows must be 16B-aligned with unit d-stride");
+  TORCH_CHECK(k.strides() == q.strides() && v.strides() == q.strides() &&
+                  k.sizes() == q.sizes() && v.sizes() == q.sizes(),
+              "q/k/v must share shape and strides");
+}
+
 void fa_fwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o,
             at::Tensor lse, int64_t N, double scale, bool causal) {
-  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16 &&
-                  q.is_contiguous(),
-              "q must be contiguous bf16");
-  const int64_t BH = q.numel() / 64 / N;
+  fa_check_qkv(q, k, v);
+  TORCH_CHECK(o.is_contiguous(), "o must be contiguous");
+  const int64_t BH = q.size(0) * q.size(1);
   throw_on(ps_fa_fwd(cur_stream(q), q.data_ptr(), k.data_ptr(), v.data_ptr(),
                      o.data_ptr(), lse.data_ptr<float>(), BH, N, (float)scale,
-                     causal ? 1 : 0),
+                     causal ? 1 : 0, q.size(1), q.stride(0), q.stride(1),
+                     q.stride(2)),
            "fa_fwd");
 }
 
@@ -491,11 +508,17 @@ void fa_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o,
             at::Tensor dout, at::Tensor lse, at::Tensor delta, at::Tensor dq,
             at::Tensor dk, at::Tensor dv, int64_t N, double scale,
             bool causal) {
-  const int64_t BH = q.numel() / 64 / N;
+  fa_check_qkv(q, k, v);
+  TORCH_CHECK(o.is_contiguous() && dout.is_contiguous() &&
+                  dq.is_contiguous() && dk.is_contiguous() &&
+                  dv.is_contiguous(),
+              "o/dout/dq/dk/dv must be contiguous");
+  const int64_t BH = q.size(0) * q.size(1);
   throw_on(ps_fa_bwd(cur_stream(q), q.data_ptr(), k.data_ptr(), v.data_ptr(),
                      o.data_ptr(), dout.data_ptr(), lse.data_ptr<float>(),
                      delta.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
-                     dv.data_ptr(), BH, N, (float)scale, causal ? 1 : 0),
+                     dv.data_ptr(), BH, N, (float)scale, causal ? 1 : 0,
+                     q.size(1), q.stride(0), q.stride(1), q.stride(2)),
            "fa_bwd");
 }
 

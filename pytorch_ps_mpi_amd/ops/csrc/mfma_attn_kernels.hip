@@ -69,10 +69,13 @@ __device__ __forceinline__ int fa_swg(int row, int g8, int stride, int mask) {
 }
 
 // 8 consecutive bf16 of row `row` at element offset `off` (row-guarded).
+// sN = element stride between consecutive sequence rows (64 if contiguous;
+// e.g. 3*H*64 for q/k/v slices of a fused qkv projection - no .contiguous()
+// copies on the model path).
 __device__ __forceinline__ bf16x8v fa_ldrow8(const __hip_bfloat16* base,
                                              int64_t row, int64_t nrows,
-                                             int off) {
-  if (row < nrows) return *(const bf16x8v*)(base + row * FA_D + off);
+                                             int off, int64_t sN) {
+  if (row < nrows) return *(const bf16x8v*)(base + row * sN + off);
   return fa_zero8();
 }
 
@@ -80,12 +83,12 @@ __device__ __forceinline__ bf16x8v fa_ldrow8(const __hip_bfloat16* base,
 // dstT[64][64] (row stride FA_BN, mask 7).  256 threads, 16 elems each.
 __device__ __forceinline__ void fa_fill_t(const __hip_bfloat16* src,
                                           int64_t r0, int64_t nrows,
-                                          __hip_bfloat16* dstT) {
+                                          __hip_bfloat16* dstT, int64_t sN) {
   const int k = threadIdx.x >> 2;         // 0..63 source row in tile
   const int d0 = (threadIdx.x & 3) * 16;  // 0..48
 #pragma unroll
   for (int h = 0; h < 2; ++h) {
-    const bf16x8v v = fa_ldrow8(src, r0 + k, nrows, d0 + h * 8);
+    const bf16x8v v = fa_ldrow8(src, r0 + k, nrows, d0 + h * 8, sN);
 #pragma unroll
     for (int j = 0; j < 8; ++j)
       dstT[fa_swe(d0 + h * 8 + j, k, FA_BN, 7)] = (__hip_bfloat16)(float)v[j];
@@ -96,10 +99,11 @@ __device__ __forceinline__ void fa_fill_t(const __hip_bfloat16* src,
 // dstT[64][32] (row stride FA_BNB, mask 3)
 __device__ __forceinline__ void fa_fill_t32(const __hip_bfloat16* src,
                                             int64_t r0, int64_t nrows,
-                                            __hip_bfloat16* dstT) {
+                                            __hip_bfloat16* dstT,
+                                            int64_t sN) {
   const int k = threadIdx.x >> 3;        // 0..31 source row in tile
   const int d0 = (threadIdx.x & 7) * 8;  // 0..56
-  const bf16x8v v = fa_ldrow8(src, r0 + k, nrows, d0);
+  const bf16x8v v = fa_ldrow8(src, r0 + k, nrows, d0, sN);
 #pragma unroll
   for (int j = 0; j < 8; ++j)
     dstT[fa_swe(d0 + j, k, FA_BNB, 3)] = (__hip_bfloat16)(float)v[j];
@@ -114,7 +118,8 @@ k_fa_fwd(const __hip_bfloat16* __restrict__ q,
          const __hip_bfloat16* __restrict__ k,
          const __hip_bfloat16* __restrict__ v,
          __hip_bfloat16* __restrict__ o, float* __restrict__ lse,
-         int64_t N, float scale, int causal) {
+         int64_t N, float scale, int causal,
+         int64_t H, int64_t sB, int64_t sH, int64_t sN) {
   // lds: vt double buffer [2][64][64] + per-wave P [4][32][64]
   __shared__ __align__(16) __hip_bfloat16 lds[2 * FA_D * FA_BN
                                               + 4 * 32 * FA_BN];
@@ -125,9 +130,9 @@ k_fa_fwd(const __hip_bfloat16* __restrict__ q,
   const int64_t bh = blockIdx.y;
   const int64_t m0 = (int64_t)blockIdx.x * FA_BM;
   const int64_t row0 = m0 + wv * 32;
-  const __hip_bfloat16* qb = q + bh * N * FA_D;
-  const __hip_bfloat16* kb = k + bh * N * FA_D;
-  const __hip_bfloat16* vb = v + bh * N * FA_D;
+  const __hip_bfloat16* qb = q + (bh / H) * sB + (bh % H) * sH;
+  const __hip_bfloat16* kb = k + (bh / H) * sB + (bh % H) * sH;
+  const __hip_bfloat16* vb = v + (bh / H) * sB + (bh % H) * sH;
   __hip_bfloat16* pbuf = lds + 2 * FA_D * FA_BN + wv * 32 * FA_BN;
 
   bf16x8v aQ[2][2];
@@ -135,7 +140,7 @@ k_fa_fwd(const __hip_bfloat16* __restrict__ q,
   for (int rf = 0; rf < 2; ++rf)
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk)
-      aQ[rf][kk] = fa_ldrow8(qb, row0 + rf * 16 + c, N, kk * 32 + g * 8);
+      aQ[rf][kk] = fa_ldrow8(qb, row0 + rf * 16 + c, N, kk * 32 + g * 8, sN);
 
   float m[2][4], l[2][4];
   f32x4v acc[2][4];
@@ -154,12 +159,13 @@ k_fa_fwd(const __hip_bfloat16* __restrict__ q,
   const int64_t kend = causal ? min(N, m0 + FA_BM) : N;
   const int64_t ntiles = (kend + FA_BN - 1) / FA_BN;
   if (ntiles <= 0) return;
-  fa_fill_t(vb, 0, N, lds);  // V tile 0 -> buffer 0
+  fa_fill_t(vb, 0, N, lds, sN);  // V tile 0 -> buffer 0
   for (int64_t t = 0; t < ntiles; ++t) {
     const int64_t k0 = t * FA_BN;
     __syncthreads();  // vt[t&1] filled; prior tile's reads complete
     if (t + 1 < ntiles)
-      fa_fill_t(vb, (t + 1) * FA_BN, N, lds + ((t + 1) & 1) * FA_D * FA_BN);
+      fa_fill_t(vb, (t + 1) * FA_BN, N,
+                lds + ((t + 1) & 1) * FA_D * FA_BN, sN);
     const __hip_bfloat16* vt = lds + (t & 1) * FA_D * FA_BN;
     if (causal && k0 > row0 + 31) continue;  // above this wave's diagonal
 
@@ -171,7 +177,7 @@ k_fa_fwd(const __hip_bfloat16* __restrict__ q,
     for (int jf = 0; jf < 4; ++jf)
 #pragma unroll
       for (int kk = 0; kk < 2; ++kk)
-        bK[jf][kk] = fa_ldrow8(kb, k0 + jf * 16 + c, N, kk * 32 + g * 8);
+        bK[jf][kk] = fa_ldrow8(kb, k0 + jf * 16 + c, N, kk * 32 + g * 8, sN);
     f32x4v S[2][4];
 #pragma unroll
     for (int rf = 0; rf < 2; ++rf)
@@ -304,7 +310,7 @@ k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
             const __hip_bfloat16* __restrict__ dout,
             const float* __restrict__ lse, const float* __restrict__ delta,
             __hip_bfloat16* __restrict__ dq, int64_t N, float scale,
-            int causal) {
+            int causal, int64_t H, int64_t sB, int64_t sH, int64_t sN) {
   // lds: Kt double buffer [2][64][32] + per-wave dS [4][32][32]
   __shared__ __align__(16) __hip_bfloat16 lds[2 * FA_D * FA_BNB
                                               + 4 * 32 * FA_BNB];
@@ -315,10 +321,10 @@ k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
   const int64_t bh = blockIdx.y;
   const int64_t m0 = (int64_t)blockIdx.x * FA_BM;
   const int64_t row0 = m0 + wv * 32;
-  const __hip_bfloat16* qb = q + bh * N * FA_D;
-  const __hip_bfloat16* kb = k + bh * N * FA_D;
-  const __hip_bfloat16* vb = v + bh * N * FA_D;
-  const __hip_bfloat16* dob = dout + bh * N * FA_D;
+  const __hip_bfloat16* qb = q + (bh / H) * sB + (bh % H) * sH;
+  const __hip_bfloat16* kb = k + (bh / H) * sB + (bh % H) * sH;
+  const __hip_bfloat16* vb = v + (bh / H) * sB + (bh % H) * sH;
+  const __hip_bfloat16* dob = dout + bh * N * FA_D;  // contiguous
   __hip_bfloat16* sbuf = lds + 2 * FA_D * FA_BNB + wv * 32 * FA_BNB;
 
   bf16x8v aQ[2][2], aDO[2][2];
@@ -327,8 +333,8 @@ k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
   for (int rf = 0; rf < 2; ++rf) {
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      aQ[rf][kk] = fa_ldrow8(qb, row0 + rf * 16 + c, N, kk * 32 + g * 8);
-      aDO[rf][kk] = fa_ldrow8(dob, row0 + rf * 16 + c, N, kk * 32 + g * 8);
+      aQ[rf][kk] = fa_ldrow8(qb, row0 + rf * 16 + c, N, kk * 32 + g * 8, sN);
+      aDO[rf][kk] = fa_ldrow8(dob, row0 + rf * 16 + c, N, kk * 32 + g * 8, FA_D);
     }
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
@@ -346,13 +352,13 @@ k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
   const int64_t kend = causal ? min(N, m0 + FA_BM) : N;
   const int64_t ntiles = (kend + FA_BNB - 1) / FA_BNB;
   if (ntiles <= 0) return;
-  fa_fill_t32(kb, 0, N, lds);
+  fa_fill_t32(kb, 0, N, lds, sN);
   for (int64_t t = 0; t < ntiles; ++t) {
     const int64_t k0 = t * FA_BNB;
     __syncthreads();
     if (t + 1 < ntiles)
       fa_fill_t32(kb, (t + 1) * FA_BNB,
-                  N, lds + ((t + 1) & 1) * FA_D * FA_BNB);
+                  N, lds + ((t + 1) & 1) * FA_D * FA_BNB, sN);
     const __hip_bfloat16* kt = lds + (t & 1) * FA_D * FA_BNB;
     if (causal && k0 > row0 + 31) continue;
 
@@ -362,8 +368,8 @@ k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
     for (int jf = 0; jf < 2; ++jf)
 #pragma unroll
       for (int kk = 0; kk < 2; ++kk) {
-        bK[jf][kk] = fa_ldrow8(kb, k0 + jf * 16 + c, N, kk * 32 + g * 8);
-        bV[jf][kk] = fa_ldrow8(vb, k0 + jf * 16 + c, N, kk * 32 + g * 8);
+        bK[jf][kk] = fa_ldrow8(kb, k0 + jf * 16 + c, N, kk * 32 + g * 8, sN);
+        bV[jf][kk] = fa_ldrow8(vb, k0 + jf * 16 + c, N, kk * 32 + g * 8, sN);
       }
     f32x4v S[2][2], dP[2][2];
 #pragma unroll
@@ -441,7 +447,8 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
              const __hip_bfloat16* __restrict__ dout,
              const float* __restrict__ lse, const float* __restrict__ delta,
              __hip_bfloat16* __restrict__ dk, __hip_bfloat16* __restrict__ dv,
-             int64_t N, float scale, int causal) {
+             int64_t N, float scale, int causal,
+             int64_t H, int64_t sB, int64_t sH, int64_t sN) {
   // lds: double-buffered {Qt,dOt} [2][2][64][32] + per-wave P/dS [4][32][32]
   __shared__ __align__(16) __hip_bfloat16 lds[4 * FA_D * FA_BNB
                                               + 4 * 32 * FA_BNB];
@@ -452,10 +459,10 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
   const int64_t bh = blockIdx.y;
   const int64_t m0 = (int64_t)blockIdx.x * FA_BM;  // first KEY of block
   const int64_t key0 = m0 + wv * 32;               // wave's first key
-  const __hip_bfloat16* qb = q + bh * N * FA_D;
-  const __hip_bfloat16* kb = k + bh * N * FA_D;
-  const __hip_bfloat16* vb = v + bh * N * FA_D;
-  const __hip_bfloat16* dob = dout + bh * N * FA_D;
+  const __hip_bfloat16* qb = q + (bh / H) * sB + (bh % H) * sH;
+  const __hip_bfloat16* kb = k + (bh / H) * sB + (bh % H) * sH;
+  const __hip_bfloat16* vb = v + (bh / H) * sB + (bh % H) * sH;
+  const __hip_bfloat16* dob = dout + bh * N * FA_D;  // contiguous
   // buffer b: qt at lds + b*2*IMG, dot right after (IMG = 64*32)
   __hip_bfloat16* sbuf = lds + 4 * FA_D * FA_BNB + wv * 32 * FA_BNB;
 
@@ -464,8 +471,8 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
   for (int rf = 0; rf < 2; ++rf)
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      aK[rf][kk] = fa_ldrow8(kb, key0 + rf * 16 + c, N, kk * 32 + g * 8);
-      aV[rf][kk] = fa_ldrow8(vb, key0 + rf * 16 + c, N, kk * 32 + g * 8);
+      aK[rf][kk] = fa_ldrow8(kb, key0 + rf * 16 + c, N, kk * 32 + g * 8, sN);
+      aV[rf][kk] = fa_ldrow8(vb, key0 + rf * 16 + c, N, kk * 32 + g * 8, sN);
     }
   f32x4v dK[2][4], dV[2][4];
 #pragma unroll
@@ -479,15 +486,15 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
   const int64_t q0_start = causal ? m0 : 0;
   const int64_t ntiles = (N - q0_start + FA_BNB - 1) / FA_BNB;
   if (ntiles <= 0) return;
-  fa_fill_t32(qb, q0_start, N, lds);
-  fa_fill_t32(dob, q0_start, N, lds + FA_D * FA_BNB);
+  fa_fill_t32(qb, q0_start, N, lds, sN);
+  fa_fill_t32(dob, q0_start, N, lds + FA_D * FA_BNB, FA_D);
   for (int64_t t = 0; t < ntiles; ++t) {
     const int64_t q0 = q0_start + t * FA_BNB;
     __syncthreads();  // buffer t&1 filled; prior tile's reads complete
     if (t + 1 < ntiles) {
       __hip_bfloat16* nb = lds + ((t + 1) & 1) * 2 * FA_D * FA_BNB;
-      fa_fill_t32(qb, q0 + FA_BNB, N, nb);
-      fa_fill_t32(dob, q0 + FA_BNB, N, nb + FA_D * FA_BNB);
+      fa_fill_t32(qb, q0 + FA_BNB, N, nb, sN);
+      fa_fill_t32(dob, q0 + FA_BNB, N, nb + FA_D * FA_BNB, FA_D);
     }
     const __hip_bfloat16* qt = lds + (t & 1) * 2 * FA_D * FA_BNB;
     const __hip_bfloat16* dot = qt + FA_D * FA_BNB;
@@ -507,8 +514,8 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
     for (int jf = 0; jf < 2; ++jf)
 #pragma unroll
       for (int kk = 0; kk < 2; ++kk) {
-        bQ[jf][kk] = fa_ldrow8(qb, q0 + jf * 16 + c, N, kk * 32 + g * 8);
-        bDO[jf][kk] = fa_ldrow8(dob, q0 + jf * 16 + c, N, kk * 32 + g * 8);
+        bQ[jf][kk] = fa_ldrow8(qb, q0 + jf * 16 + c, N, kk * 32 + g * 8, sN);
+        bDO[jf][kk] = fa_ldrow8(dob, q0 + jf * 16 + c, N, kk * 32 + g * 8, FA_D);
       }
     f32x4v St[2][2], dPt[2][2];
 #pragma unroll
@@ -632,20 +639,21 @@ extern "C" {
 
 int ps_fa_fwd(void* stream_, const void* q, const void* k, const void* v,
               void* o, float* lse, int64_t BH, int64_t N, float scale,
-              int causal) {
+              int causal, int64_t H, int64_t sB, int64_t sH, int64_t sN) {
   hipStream_t s = (hipStream_t)stream_;
   dim3 grid((unsigned)((N + FA_BM - 1) / FA_BM), (unsigned)BH);
   hipLaunchKernelGGL(k_fa_fwd, grid, dim3(256), 0, s,
                      (const __hip_bfloat16*)q, (const __hip_bfloat16*)k,
                      (const __hip_bfloat16*)v, (__hip_bfloat16*)o, lse, N,
-                     scale, causal);
+                     scale, causal, H, sB, sH, sN);
   return (int)hipGetLastError();
 }
 
 int ps_fa_bwd(void* stream_, const void* q, const void* k, const void* v,
               const void* o, const void* dout, const float* lse, float* delta,
               void* dq, void* dk, void* dv, int64_t BH, int64_t N,
-              float scale, int causal) {
+              float scale, int causal, int64_t H, int64_t sB, int64_t sH,
+              int64_t sN) {
   hipStream_t s = (hipStream_t)stream_;
   const int64_t rows = BH * N;
   int64_t db = (rows + 3) / 4;
@@ -657,12 +665,13 @@ int ps_fa_bwd(void* stream_, const void* q, const void* k, const void* v,
   hipLaunchKernelGGL(k_fa_bwd_dq, grid, dim3(256), 0, s,
                      (const __hip_bfloat16*)q, (const __hip_bfloat16*)k,
                      (const __hip_bfloat16*)v, (const __hip_bfloat16*)dout,
-                     lse, delta, (__hip_bfloat16*)dq, N, scale, causal);
+                     lse, delta, (__hip_bfloat16*)dq, N, scale, causal,
+                     H, sB, sH, sN);
   hipLaunchKernelGGL(k_fa_bwd_dkv, grid, dim3(256), 0, s,
                      (const __hip_bfloat16*)q, (const __hip_bfloat16*)k,
                      (const __hip_bfloat16*)v, (const __hip_bfloat16*)dout,
                      lse, delta, (__hip_bfloat16*)dk, (__hip_bfloat16*)dv, N,
-                     scale, causal);
+                     scale, causal, H, sB, sH, sN);
   return (int)hipGetLastError();
 }
 

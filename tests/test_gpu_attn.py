@@ -88,3 +88,23 @@ def test_fused_sdpa_fallback_d128():
     q = torch.randn(1, 2, 16, 128, device=dev, dtype=torch.bfloat16)
     o = fused_sdpa(q, q, q)
     assert torch.isfinite(o.float()).all()
+
+
+def test_fused_sdpa_strided_qkv(monkeypatch):
+    """Head-slices of a fused qkv projection (the model path) hit the MFMA
+    kernels WITHOUT .contiguous() copies — strides are consumed natively."""
+    monkeypatch.setenv("PS_AMD_ATTN", "mfma")
+    torch.manual_seed(0)
+    dev = "cuda:0"
+    B, H, N, D = 2, 3, 197, 64
+    qkv = torch.randn(B, N, 3, H, D, device=dev).bfloat16()
+    q, k, v = qkv.permute(2, 0, 3, 1, 4)  # [B,H,N,D] non-contiguous views
+    assert not q.is_contiguous()
+    ref = F.scaled_dot_product_attention(q.float(), k.float(), v.float())
+    o = fused_sdpa(q, k, v)
+    assert (o.float() - ref).abs().max().item() < 0.05
+    # causal too
+    ref = F.scaled_dot_product_attention(q.float(), k.float(), v.float(),
+                                         is_causal=True)
+    o = fused_sdpa(q, k, v, is_causal=True)
+    assert (o.float() - ref).abs().max().item() < 0.05
