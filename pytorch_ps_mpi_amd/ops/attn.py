@@ -32,7 +32,9 @@ class _FusedSDPA(torch.autograd.Function):
     def forward(ctx, q, k, v, causal, use_mfma):
         B, H, N, D = q.shape
         scale = 1.0 / math.sqrt(D)
-        o = torch.empty_like(q)
+        # explicit shape: empty_like would inherit q's (possibly strided)
+        # layout, and o is always written contiguous
+        o = torch.empty((B, H, N, D), dtype=q.dtype, device=q.device)
         lse = torch.empty(B * H * N, dtype=torch.float32, device=q.device)
         if use_mfma:
             _EXT.fa_fwd(q, k, v, o, lse, N, scale, causal)
@@ -50,9 +52,9 @@ class _FusedSDPA(torch.autograd.Function):
         scale = 1.0 / math.sqrt(D)
         dout = dout.contiguous()
         delta = torch.empty_like(lse)
-        dq = torch.empty_like(q)
-        dk = torch.empty_like(k)
-        dv = torch.empty_like(v)
+        dq = torch.empty((B, H, N, D), dtype=q.dtype, device=q.device)
+        dk = torch.empty_like(dq)
+        dv = torch.empty_like(dq)
         if ctx.use_mfma:
             _EXT.fa_bwd(q, k, v, o, dout, lse, delta, dq, dk, dv, N, scale,
                         ctx.causal)
