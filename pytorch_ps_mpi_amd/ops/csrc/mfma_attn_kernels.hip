@@ -39,7 +39,10 @@ typedef __attribute__((ext_vector_type(4))) float f32x4v;
 #define MFMA16(A, B, C) __builtin_amdgcn_mfma_f32_16x16x32_bf16(A, B, C, 0, 0, 0)
 
 #define FA_D 64
-#define FA_BM 128   // rows (fwd/dq) or keys (dkv) per 4-wave block
+#define FA_BM 256   // rows (fwd/dq) or keys (dkv) per 8-wave (512-thr) block
+                    // (256 not 128: each kv/q tile is re-read once per block,
+                    // so doubling the block halves the cross-block HBM/L3
+                    // re-read traffic the bwd kernels are bound by in-model)
 #define FA_BN 64    // fwd kv tile width
 #define FA_BNB 32   // bwd tile width: the bwd kernels carry 2-3x the live
                     // state of fwd (two C-tiles + two accumulators), and at
@@ -84,15 +87,12 @@ __device__ __forceinline__ bf16x8v fa_ldrow8(const __hip_bfloat16* base,
 __device__ __forceinline__ void fa_fill_t(const __hip_bfloat16* src,
                                           int64_t r0, int64_t nrows,
                                           __hip_bfloat16* dstT, int64_t sN) {
-  const int k = threadIdx.x >> 2;         // 0..63 source row in tile
-  const int d0 = (threadIdx.x & 3) * 16;  // 0..48
+  const int k = threadIdx.x >> 3;        // 0..63 source row (512 threads)
+  const int d0 = (threadIdx.x & 7) * 8;  // 0..56
+  const bf16x8v v = fa_ldrow8(src, r0 + k, nrows, d0, sN);
 #pragma unroll
-  for (int h = 0; h < 2; ++h) {
-    const bf16x8v v = fa_ldrow8(src, r0 + k, nrows, d0 + h * 8, sN);
-#pragma unroll
-    for (int j = 0; j < 8; ++j)
-      dstT[fa_swe(d0 + h * 8 + j, k, FA_BN, 7)] = (__hip_bfloat16)(float)v[j];
-  }
+  for (int j = 0; j < 8; ++j)
+    dstT[fa_swe(d0 + j, k, FA_BN, 7)] = (__hip_bfloat16)(float)v[j];
 }
 
 // 32-row variant for the bwd kernels: rows [r0, r0+32) -> swizzled
@@ -101,6 +101,7 @@ __device__ __forceinline__ void fa_fill_t32(const __hip_bfloat16* src,
                                             int64_t r0, int64_t nrows,
                                             __hip_bfloat16* dstT,
                                             int64_t sN) {
+  if (threadIdx.x >= 256) return;        // 32x64 tile: first 256 threads
   const int k = threadIdx.x >> 3;        // 0..31 source row in tile
   const int d0 = (threadIdx.x & 7) * 8;  // 0..56
   const bf16x8v v = fa_ldrow8(src, r0 + k, nrows, d0, sN);
@@ -113,16 +114,16 @@ __device__ __forceinline__ void fa_fill_t32(const __hip_bfloat16* src,
 // forward
 // ---------------------------------------------------------------------------
 
-__global__ void __launch_bounds__(256)
+__global__ void __launch_bounds__(512)
 k_fa_fwd(const __hip_bfloat16* __restrict__ q,
          const __hip_bfloat16* __restrict__ k,
          const __hip_bfloat16* __restrict__ v,
          __hip_bfloat16* __restrict__ o, float* __restrict__ lse,
          int64_t N, float scale, int causal,
          int64_t H, int64_t sB, int64_t sH, int64_t sN) {
-  // lds: vt double buffer [2][64][64] + per-wave P [4][32][64]
+  // lds: vt double buffer [2][64][64] + per-wave P [8][32][64]
   __shared__ __align__(16) __hip_bfloat16 lds[2 * FA_D * FA_BN
-                                              + 4 * 32 * FA_BN];
+                                              + 8 * 32 * FA_BN];
   const int lane = threadIdx.x & 63;
   const int wv = threadIdx.x >> 6;
   const int c = lane & 15;
@@ -303,7 +304,7 @@ k_fa_delta(const __hip_bfloat16* __restrict__ dout,
   }
 }
 
-__global__ void __launch_bounds__(256)
+__global__ void __launch_bounds__(512)
 k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
             const __hip_bfloat16* __restrict__ k,
             const __hip_bfloat16* __restrict__ v,
@@ -311,9 +312,9 @@ k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
             const float* __restrict__ lse, const float* __restrict__ delta,
             __hip_bfloat16* __restrict__ dq, int64_t N, float scale,
             int causal, int64_t H, int64_t sB, int64_t sH, int64_t sN) {
-  // lds: Kt double buffer [2][64][32] + per-wave dS [4][32][32]
+  // lds: Kt double buffer [2][64][32] + per-wave dS [8][32][32]
   __shared__ __align__(16) __hip_bfloat16 lds[2 * FA_D * FA_BNB
-                                              + 4 * 32 * FA_BNB];
+                                              + 8 * 32 * FA_BNB];
   const int lane = threadIdx.x & 63;
   const int wv = threadIdx.x >> 6;
   const int c = lane & 15;
@@ -440,7 +441,7 @@ k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
     }
 }
 
-__global__ void __launch_bounds__(256)
+__global__ void __launch_bounds__(512)
 k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
              const __hip_bfloat16* __restrict__ k,
              const __hip_bfloat16* __restrict__ v,
@@ -449,9 +450,9 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
              __hip_bfloat16* __restrict__ dk, __hip_bfloat16* __restrict__ dv,
              int64_t N, float scale, int causal,
              int64_t H, int64_t sB, int64_t sH, int64_t sN) {
-  // lds: double-buffered {Qt,dOt} [2][2][64][32] + per-wave P/dS [4][32][32]
+  // lds: double-buffered {Qt,dOt} [2][2][64][32] + per-wave P/dS [8][32][32]
   __shared__ __align__(16) __hip_bfloat16 lds[4 * FA_D * FA_BNB
-                                              + 4 * 32 * FA_BNB];
+                                              + 8 * 32 * FA_BNB];
   const int lane = threadIdx.x & 63;
   const int wv = threadIdx.x >> 6;
   const int c = lane & 15;
@@ -642,7 +643,7 @@ int ps_fa_fwd(void* stream_, const void* q, const void* k, const void* v,
               int causal, int64_t H, int64_t sB, int64_t sH, int64_t sN) {
   hipStream_t s = (hipStream_t)stream_;
   dim3 grid((unsigned)((N + FA_BM - 1) / FA_BM), (unsigned)BH);
-  hipLaunchKernelGGL(k_fa_fwd, grid, dim3(256), 0, s,
+  hipLaunchKernelGGL(k_fa_fwd, grid, dim3(512), 0, s,
                      (const __hip_bfloat16*)q, (const __hip_bfloat16*)k,
                      (const __hip_bfloat16*)v, (__hip_bfloat16*)o, lse, N,
                      scale, causal, H, sB, sH, sN);
@@ -662,12 +663,12 @@ int ps_fa_bwd(void* stream_, const void* q, const void* k, const void* v,
                      (const __hip_bfloat16*)dout, (const __hip_bfloat16*)o,
                      delta, rows);
   dim3 grid((unsigned)((N + FA_BM - 1) / FA_BM), (unsigned)BH);
-  hipLaunchKernelGGL(k_fa_bwd_dq, grid, dim3(256), 0, s,
+  hipLaunchKernelGGL(k_fa_bwd_dq, grid, dim3(512), 0, s,
                      (const __hip_bfloat16*)q, (const __hip_bfloat16*)k,
                      (const __hip_bfloat16*)v, (const __hip_bfloat16*)dout,
                      lse, delta, (__hip_bfloat16*)dq, N, scale, causal,
                      H, sB, sH, sN);
-  hipLaunchKernelGGL(k_fa_bwd_dkv, grid, dim3(256), 0, s,
+  hipLaunchKernelGGL(k_fa_bwd_dkv, grid, dim3(512), 0, s,
                      (const __hip_bfloat16*)q, (const __hip_bfloat16*)k,
                      (const __hip_bfloat16*)v, (const __hip_bfloat16*)dout,
                      lse, delta, (__hip_bfloat16*)dk, (__hip_bfloat16*)dv, N,
