@@ -34,8 +34,8 @@ class _FusedBN(torch.autograd.Function):
         scale = torch.empty(C, **f32)
         shift = torch.empty(C, **f32)
         if training:
-            psum = torch.empty(C, **f32)
-            psumsq = torch.empty(C, **f32)
+            stats = torch.empty(2 * C, **f32)  # contiguous: one fill launch
+            psum, psumsq = stats[:C], stats[C:]
             _EXT.bn_fwd_stats(x, psum, psumsq, rows, C)
             mean = torch.empty(C, **f32)
             invstd = torch.empty(C, **f32)
@@ -66,8 +66,8 @@ class _FusedBN(torch.autograd.Function):
             dy = dy.contiguous(memory_format=_CL)
         dev = x.device
         f32 = dict(dtype=torch.float32, device=dev)
-        dsum = torch.empty(C, **f32)
-        dxsum = torch.empty(C, **f32)
+        dstats = torch.empty(2 * C, **f32)  # contiguous: one fill launch
+        dsum, dxsum = dstats[:C], dstats[C:]
         _EXT.bn_bwd_stats(x, dy, z, scale, shift, dsum, dxsum, rows, C,
                           ctx.relu)
         wd = weight.dtype if weight is not None else x.dtype

@@ -320,7 +320,11 @@ k_bn_bwd_coef(const float* __restrict__ dsum,     // sum(dy_eff)
 static inline dim3 bn_reduce_grid(int64_t rows, int64_t C) {
   const int ct = (int)(C / BN_CT);
   int64_t yb = (rows + 31) / 32;
-  int64_t cap = 4096 / ct;
+  // cap total blocks at 1024 (4096 waves fills the chip): every y-block
+  // costs one serialized atomicAdd chain per channel word, and at the old
+  // 4096-block cap the per-word chain (~11ns/atomic) was ~45us per C=64
+  // reduce
+  int64_t cap = 1024 / ct;
   if (cap < 1) cap = 1;
   if (yb > cap) yb = cap;
   return dim3(ct, (unsigned)yb);
@@ -347,10 +351,16 @@ extern "C" {
 int ps_bn_fwd_stats(void* stream_, const void* x, float* psum, float* psumsq,
                     int64_t rows, int64_t C) {
   hipStream_t s = (hipStream_t)stream_;
-  hipError_t e = hipMemsetAsync(psum, 0, C * sizeof(float), s);
-  if (e) return (int)e;
-  e = hipMemsetAsync(psumsq, 0, C * sizeof(float), s);
-  if (e) return (int)e;
+  hipError_t e;
+  if (psumsq == psum + C) {  // contiguous pair: one fill launch
+    e = hipMemsetAsync(psum, 0, 2 * C * sizeof(float), s);
+    if (e) return (int)e;
+  } else {
+    e = hipMemsetAsync(psum, 0, C * sizeof(float), s);
+    if (e) return (int)e;
+    e = hipMemsetAsync(psumsq, 0, C * sizeof(float), s);
+    if (e) return (int)e;
+  }
   hipLaunchKernelGGL((k_bn_reduce<false, false>), bn_reduce_grid(rows, C),
                      dim3(BN_BLOCK), 0, s, (const __hip_bfloat16*)x, nullptr,
                      nullptr, nullptr, nullptr, psum, psumsq, rows, C, 0);
@@ -410,10 +420,16 @@ int ps_bn_bwd_stats(void* stream_, const void* x, const void* dy,
                     float* dsum, float* dxsum, int64_t rows, int64_t C,
                     int relu) {
   hipStream_t s = (hipStream_t)stream_;
-  hipError_t e = hipMemsetAsync(dsum, 0, C * sizeof(float), s);
-  if (e) return (int)e;
-  e = hipMemsetAsync(dxsum, 0, C * sizeof(float), s);
-  if (e) return (int)e;
+  hipError_t e;
+  if (dxsum == dsum + C) {
+    e = hipMemsetAsync(dsum, 0, 2 * C * sizeof(float), s);
+    if (e) return (int)e;
+  } else {
+    e = hipMemsetAsync(dsum, 0, C * sizeof(float), s);
+    if (e) return (int)e;
+    e = hipMemsetAsync(dxsum, 0, C * sizeof(float), s);
+    if (e) return (int)e;
+  }
   if (z != nullptr)
     hipLaunchKernelGGL((k_bn_reduce<true, true>), bn_reduce_grid(rows, C),
                        dim3(BN_BLOCK), 0, s, (const __hip_bfloat16*)x,
