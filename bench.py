@@ -38,7 +38,10 @@ def parse_args():
     ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
     ap.add_argument("--optim", default="sgd", choices=["sgd", "adam"])
     ap.add_argument("--lr", type=float, default=0.0125)
-    ap.add_argument("--bucket-mb", type=float, default=50.0)
+    ap.add_argument("--bucket-mb", type=float, default=None,
+                    help="bucket size; default 8 MB for async (ResNet-50 -> "
+                    "~7 buckets so hook-pipelined pushes overlap backward, "
+                    "docs/DESIGN.md scaling model), 50 MB otherwise")
     ap.add_argument("--dedicated-ps", action="store_true")
     ap.add_argument("--window", type=int, default=2)
     ap.add_argument("--max-stale", type=int, default=8)
@@ -83,6 +86,8 @@ def main():
                      and args.model != "vit_b16")
     if channels_last:
         model = model.to(memory_format=torch.channels_last)
+    if args.bucket_mb is None:
+        args.bucket_mb = 8.0 if args.mode == "async" else 50.0
     opt_cls = {"sgd": SGD, "adam": Adam}[args.optim]
     opt_kw = dict(mode=args.mode, code=args.codec,
                   bucket_mb=args.bucket_mb, grad_scale="mean",
