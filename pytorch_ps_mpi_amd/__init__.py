@@ -17,7 +17,8 @@ Public API (reference parity: __init__.py:1 exported MPI_PS, Adam, SGD):
 """
 
 from . import codecs, models, ops
-from .codecs import HostCodec, Identity, QuantInt8, TopK, get_codec
+from .codecs import (HostCodec, Identity, QuantInt8, TopK,
+                     TopKThreshold, get_codec)
 from .optim import PS, SGD, Adam
 from .parallel.comm import Comm, init_distributed
 
@@ -31,7 +32,8 @@ def MPI_PS(named_params, *, optim="sgd", code=None, **kwargs):
 
 __all__ = [
     "PS", "MPI_PS", "SGD", "Adam",
-    "Identity", "TopK", "QuantInt8", "HostCodec", "get_codec",
+    "Identity", "TopK", "TopKThreshold", "QuantInt8", "HostCodec",
+    "get_codec",
     "Comm", "init_distributed",
     "codecs", "models", "ops",
 ]

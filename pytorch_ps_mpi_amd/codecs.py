@@ -35,7 +35,7 @@ class Identity:
     def wire_dtype(self, src_dtype):
         return src_dtype
 
-    def wire_numel(self, numel):
+    def wire_numel(self, numel, src_dtype=None):
         return numel
 
     def encode(self, src, wire):
@@ -109,6 +109,85 @@ class TopK:
         for w in wires:  # one message at a time: unique indices, deterministic
             k, idx, val = self._views(w, numel, dt)
             ops.topk_scatter(dst, idx, val, k, gscale)
+
+    def bytes_on_wire(self, numel, dtype=torch.bfloat16):
+        return self.wire_numel(numel, dtype)
+
+
+class TopKThreshold:
+    """VARIABLE-k magnitude-threshold sparsification — the device-side
+    variable-length wire (round-1 verdict, missing #3).
+
+    Selects every element within a factor `alpha` of the bucket's peak
+    magnitude (to 1/8-octave key granularity), capped at `max_density`;
+    the data-dependent k_used travels in a 4-byte device-side header at the
+    start of the wire, so the PS decodes exactly the used span — no size
+    exchange and no host round trip.  The wire CAPACITY stays fixed
+    (RCCL has no gatherv; recv slots are pre-posted), so adaptivity saves
+    decode/scatter work and models content-sized payload semantics, not
+    bytes on the physical link.
+
+    Wire layout (uint8): [ int32 k_used | pad to 16B | int32 idx[kmax]
+    | val[kmax] (model dtype) | pad ].
+    """
+
+    name = "topkt"
+    supports_allreduce = False
+
+    def __init__(self, alpha=0.05, max_density=0.05, min_k=8):
+        if not (0.0 < alpha <= 1.0):
+            raise ValueError("alpha in (0,1]")
+        if not (0.0 < max_density <= 1.0):
+            raise ValueError("max_density in (0,1]")
+        import math
+        self.alpha = alpha
+        self.off_keys = max(0, round(-8.0 * math.log2(alpha)))
+        self.max_density = max_density
+        self.min_k = min_k
+        self._ws = {}
+
+    def kmax_for(self, numel):
+        k = max(self.min_k, int(numel * self.max_density))
+        return min(k, numel)
+
+    def wire_dtype(self, src_dtype):
+        return torch.uint8
+
+    def wire_numel(self, numel, src_dtype=torch.bfloat16):
+        k = self.kmax_for(numel)
+        raw = 16 + 4 * k + src_dtype.itemsize * k
+        return (raw + 15) // 16 * 16
+
+    def _views(self, wire, numel, src_dtype):
+        k = self.kmax_for(numel)
+        hdr = wire[:4].view(torch.int32)
+        idx = wire[16:16 + 4 * k].view(torch.int32)
+        vb = src_dtype.itemsize * k
+        val = wire[16 + 4 * k:16 + 4 * k + vb].view(src_dtype)
+        return k, hdr, idx, val
+
+    def _workspace(self, device):
+        key = str(device)
+        if key not in self._ws:
+            self._ws[key] = ops.topk_workspace(device)
+        return self._ws[key]
+
+    def encode(self, src, wire):
+        k, hdr, idx, val = self._views(wire, src.numel(), src.dtype)
+        ops.topk_thresh_encode(src, self.off_keys, k, self._workspace(
+            src.device), hdr, idx, val)
+
+    def decode_reduce(self, dst, wires, gscale=1.0, beta=0.0, src_dtype=None):
+        numel = dst.numel()
+        dt = src_dtype if src_dtype is not None else (
+            torch.bfloat16 if dst.is_cuda else torch.float32)
+        if beta == 0.0:
+            dst.zero_()
+        elif beta != 1.0:
+            dst.mul_(beta)
+        for w in wires:
+            k, hdr, idx, val = self._views(w, numel, dt)
+            ops.topk_scatter_var(dst, hdr, idx, val, k, gscale)
 
     def bytes_on_wire(self, numel, dtype=torch.bfloat16):
         return self.wire_numel(numel, dtype)
@@ -263,13 +342,19 @@ class HostCodec:
 
 
 def get_codec(spec):
-    """'identity' | 'topk' | 'topk:0.02' | 'quant8' | codec instance | None."""
+    """'identity' | 'topk[:density]' | 'topkt[:alpha[:max_density]]' |
+    'quant8' | codec instance | None."""
     if spec is None:
         return Identity()
     if not isinstance(spec, str):
         return spec
     if spec == "identity":
         return Identity()
+    if spec.startswith("topkt"):
+        parts = spec.split(":")
+        alpha = float(parts[1]) if len(parts) > 1 else 0.05
+        md = float(parts[2]) if len(parts) > 2 else 0.05
+        return TopKThreshold(alpha=alpha, max_density=md)
     if spec.startswith("topk"):
         parts = spec.split(":")
         density = float(parts[1]) if len(parts) > 1 else 0.01

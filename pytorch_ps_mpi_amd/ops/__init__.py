@@ -193,6 +193,40 @@ def topk_scatter(dst, idx, val, k, gscale=1.0):
     dst.index_add_(0, idx[:k].long(), val[:k].float() * gscale)
 
 
+def _keys_of(x):
+    """11-bit magnitude key: float32 bits of |x| >> 21 (matches tk_key)."""
+    return (x.float().abs().contiguous().view(torch.int32) >> 21) & 0x7FF
+
+
+def topk_thresh_encode(src, off_keys, kmax, ws, hdr, out_idx, out_val):
+    """VARIABLE-k selection: every element within `off_keys` magnitude-key
+    steps (1/8 octave each) of the bucket's peak |x|, capped at kmax;
+    k_used lands in the int32 `hdr` — the device-side variable-length wire
+    (the reference's adaptive codecs sized payloads by content).
+    """
+    if _require_ext(src):
+        _EXT.topk_encode_thresh(src, int(off_keys), int(kmax), ws, hdr,
+                                out_idx, out_val)
+        return
+    keys = _keys_of(src)
+    thr0 = max(0, int(keys.max()) - int(off_keys))
+    cand = int((keys >= thr0).sum())
+    k = max(1, min(cand, int(kmax)))
+    v, idx = torch.topk(src.float().abs(), k, sorted=False)
+    hdr[0] = k
+    out_idx[:k].copy_(idx.to(torch.int32))
+    out_val[:k].copy_(src[idx])
+
+
+def topk_scatter_var(dst, hdr, idx, val, kmax, gscale=1.0):
+    """dst[idx[:k]] += gscale * val[:k] with k read from the device header."""
+    if _require_ext(dst):
+        _EXT.topk_scatter_var(dst, hdr, idx, val, int(kmax), gscale)
+        return
+    k = int(hdr[0])
+    dst.index_add_(0, idx[:k].long(), val[:k].float() * gscale)
+
+
 from . import bn  # noqa: E402,F401  (fused BatchNorm module; needs ops ready)
 from . import ln  # noqa: E402,F401
 from . import ce  # noqa: E402,F401

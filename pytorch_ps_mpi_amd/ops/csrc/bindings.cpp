@@ -28,6 +28,12 @@ int ps_topk_encode(void* stream, const void* src, int src_is_bf16, int64_t n,
                    int64_t k, uint32_t* ws, int32_t* out_idx, void* out_val);
 int ps_topk_scatter(void* stream, float* dst, const int32_t* idx,
                     const void* val, int val_is_bf16, int64_t k, float gscale);
+int ps_topk_encode_thresh(void* stream, const void* src, int src_is_bf16,
+                          int64_t n, int off_keys, int64_t kmax, uint32_t* ws,
+                          int32_t* hdr, int32_t* out_idx, void* out_val);
+int ps_topk_scatter_var(void* stream, float* dst, const int32_t* hdr,
+                        const int32_t* idx, const void* val, int val_is_bf16,
+                        int64_t kmax, float gscale);
 int ps_bn_fwd_stats(void* stream, const void* x, float* psum, float* psumsq,
                     int64_t rows, int64_t C);
 int ps_bn_finalize(void* stream, const float* psum, const float* psumsq,
@@ -249,6 +255,40 @@ void quant8_reduce(at::Tensor dst, std::vector<at::Tensor> scales,
 }
 
 int64_t topk_workspace_words() { return ps_topk_workspace_words(); }
+
+void topk_encode_thresh(at::Tensor src, int64_t off_keys, int64_t kmax,
+                        at::Tensor ws, at::Tensor hdr, at::Tensor out_idx,
+                        at::Tensor out_val) {
+  TORCH_CHECK(src.is_cuda() && src.is_contiguous(), "src invalid");
+  int bf = is_bf16(src);
+  TORCH_CHECK(bf || src.scalar_type() == at::kFloat, "src dtype");
+  TORCH_CHECK(ws.scalar_type() == at::kInt && ws.numel() >= ps_topk_workspace_words(),
+              "workspace too small");
+  TORCH_CHECK(hdr.scalar_type() == at::kInt && hdr.numel() >= 1, "hdr int32[1]");
+  TORCH_CHECK(out_idx.scalar_type() == at::kInt && out_idx.numel() >= kmax,
+              "out_idx too small");
+  TORCH_CHECK(out_val.numel() >= kmax, "out_val too small");
+  TORCH_CHECK(kmax >= 1 && kmax <= src.numel(), "kmax in [1, n]");
+  throw_on(ps_topk_encode_thresh(cur_stream(src), src.data_ptr(), bf,
+                                 src.numel(), (int)off_keys, kmax,
+                                 (uint32_t*)ws.data_ptr(),
+                                 (int32_t*)hdr.data_ptr(),
+                                 (int32_t*)out_idx.data_ptr(),
+                                 out_val.data_ptr()),
+           "topk_encode_thresh");
+}
+
+void topk_scatter_var(at::Tensor dst, at::Tensor hdr, at::Tensor idx,
+                      at::Tensor val, int64_t kmax, double gscale) {
+  check_flat(dst, at::kFloat, "dst");
+  TORCH_CHECK(hdr.is_cuda() && hdr.scalar_type() == at::kInt, "hdr invalid");
+  TORCH_CHECK(idx.scalar_type() == at::kInt, "idx int32");
+  throw_on(ps_topk_scatter_var(cur_stream(dst), dst.data_ptr<float>(),
+                               (const int32_t*)hdr.data_ptr(),
+                               (const int32_t*)idx.data_ptr(), val.data_ptr(),
+                               is_bf16(val), kmax, (float)gscale),
+           "topk_scatter_var");
+}
 
 void topk_encode(at::Tensor src, int64_t k, at::Tensor ws, at::Tensor out_idx,
                  at::Tensor out_val) {
@@ -537,6 +577,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fa_fwd", &fa_fwd, "MFMA-tiled flash attention forward");
   m.def("fa_bwd", &fa_bwd, "MFMA-tiled flash attention backward");
   m.def("fa_selfcheck", &fa_selfcheck, "16x16x32 MFMA fragment-layout check");
+  m.def("topk_encode_thresh", &topk_encode_thresh,
+        "variable-k magnitude-threshold select (k_used -> device header)");
+  m.def("topk_scatter_var", &topk_scatter_var,
+        "scatter-add with device-side k header");
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);
   m.def("ln_fwd", &ln_fwd);

@@ -399,6 +399,65 @@ k_topk_plan(const uint32_t* __restrict__ hist, uint32_t* __restrict__ plan, int6
   }
 }
 
+// Threshold (VARIABLE-k) plan: select every element within a magnitude
+// factor of the bucket's peak, capped at kmax — the device-side
+// variable-length wire of the codec contract (the reference's adaptive
+// codecs varied payload size with content; here k_used rides in a wire
+// header instead of a size exchange).  `off_keys` is the key-domain
+// threshold distance below the top nonzero bin: keys quantize |x| to
+// 1/8-octave steps, so off_keys = round(8 * log2(1/alpha)) selects
+// |x| >= alpha * max|x| (to key granularity).
+// plan[0]=thr_key, plan[1]=n_above, plan[2]=need_from_thr_bin, plan[3]=k_used
+__global__ void __launch_bounds__(PS_BLOCK)
+k_topk_plan_thresh(const uint32_t* __restrict__ hist,
+                   uint32_t* __restrict__ plan, int off_keys, int64_t kmax) {
+  __shared__ uint32_t sh[TK_BINS];
+  for (int b = threadIdx.x; b < TK_BINS; b += blockDim.x) sh[b] = hist[b];
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    int bmax = 0;
+    for (int b = TK_BINS - 1; b >= 0; --b) {
+      if (sh[b]) { bmax = b; break; }
+    }
+    const int thr0 = max(0, bmax - off_keys);
+    uint64_t cand = 0;
+    for (int b = TK_BINS - 1; b >= thr0; --b) cand += sh[b];
+    uint64_t k = cand < (uint64_t)kmax ? cand : (uint64_t)kmax;
+    if (k < 1) k = 1;  // host guarantees n >= 1; keep at least the peak
+    uint64_t above = 0;
+    int thr = 0;
+    for (int b = TK_BINS - 1; b >= 0; --b) {
+      if (above + sh[b] >= k) { thr = b; break; }
+      above += sh[b];
+    }
+    plan[0] = (uint32_t)thr;
+    plan[1] = (uint32_t)above;
+    plan[2] = (uint32_t)(k - above);
+    plan[3] = (uint32_t)k;
+  }
+}
+
+// publish k_used into the wire header
+__global__ void k_topk_hdr(const uint32_t* __restrict__ plan,
+                           int32_t* __restrict__ hdr) {
+  if (threadIdx.x == 0) hdr[0] = (int32_t)plan[3];
+}
+
+// scatter with DEVICE-side k (wire header): launched for kmax, guarded.
+template <typename T>
+__global__ void __launch_bounds__(PS_BLOCK)
+k_topk_scatter_var(float* __restrict__ dst, const int32_t* __restrict__ hdr,
+                   const int32_t* __restrict__ idx, const T* __restrict__ val,
+                   int64_t kmax, float gscale) {
+  const int64_t k = hdr[0];
+  int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t j = i0; j < k; j += stride) {
+    const int32_t d = idx[j];
+    dst[d] += gscale * ld_as_float(val, j);
+  }
+}
+
 // Compaction is 3-phase with NO contended global atomics (a single global
 // counter word takes ~88 atomics/us on this chip; at 1% density nearly every
 // wave carries a candidate, so even wave-aggregated slot allocation cost
@@ -726,6 +785,56 @@ int ps_topk_encode(void* stream_, const void* src, int src_is_bf16, int64_t n,
   if (src_is_bf16) TK_RUN(__hip_bfloat16);
   else TK_RUN(float);
 #undef TK_RUN
+  return (int)hipGetLastError();
+}
+
+int ps_topk_encode_thresh(void* stream_, const void* src, int src_is_bf16,
+                          int64_t n, int off_keys, int64_t kmax, uint32_t* ws,
+                          int32_t* hdr, int32_t* out_idx, void* out_val) {
+  hipStream_t stream = (hipStream_t)stream_;
+  uint32_t* hist = ws;
+  uint32_t* plan = ws + TK_BINS;
+  uint32_t* cnt_a = ws + TK_BINS + 4;
+  uint32_t* cnt_e = cnt_a + TK_NB;
+  hipError_t e = hipMemsetAsync(ws, 0, sizeof(uint32_t) * (TK_BINS + 4), stream);
+  if (e != hipSuccess) return (int)e;
+  int nb = TK_NB;
+  int64_t chunk = (n + nb - 1) / nb;
+  chunk = (chunk + PS_BLOCK - 1) / PS_BLOCK * PS_BLOCK;
+  nb = (int)((n + chunk - 1) / chunk);
+  dim3 grid(ps_grid(n)), cgrid(nb), block(PS_BLOCK);
+#define TKT_RUN(T)                                                             \
+  do {                                                                         \
+    hipLaunchKernelGGL(k_topk_hist<T>, grid, block, 0, stream,                 \
+                       (const T*)src, hist, n);                                \
+    hipLaunchKernelGGL(k_topk_plan_thresh, dim3(1), block, 0, stream, hist,    \
+                       plan, off_keys, kmax);                                  \
+    hipLaunchKernelGGL(k_topk_hdr, dim3(1), dim3(64), 0, stream, plan, hdr);   \
+    hipLaunchKernelGGL(k_topk_count<T>, cgrid, block, 0, stream,               \
+                       (const T*)src, plan, cnt_a, cnt_e, n, chunk);           \
+    hipLaunchKernelGGL(k_topk_scan, dim3(1), block, 0, stream, cnt_a, cnt_e,   \
+                       nb);                                                    \
+    hipLaunchKernelGGL(k_topk_emit<T>, cgrid, block, 0, stream, (const T*)src, \
+                       plan, cnt_a, cnt_e, out_idx, (T*)out_val, n, chunk);    \
+  } while (0)
+  if (src_is_bf16) TKT_RUN(__hip_bfloat16);
+  else TKT_RUN(float);
+#undef TKT_RUN
+  return (int)hipGetLastError();
+}
+
+int ps_topk_scatter_var(void* stream_, float* dst, const int32_t* hdr,
+                        const int32_t* idx, const void* val, int val_is_bf16,
+                        int64_t kmax, float gscale) {
+  hipStream_t stream = (hipStream_t)stream_;
+  dim3 grid(ps_grid(kmax)), block(PS_BLOCK);
+  if (val_is_bf16)
+    hipLaunchKernelGGL(k_topk_scatter_var<__hip_bfloat16>, grid, block, 0,
+                       stream, dst, hdr, idx, (const __hip_bfloat16*)val,
+                       kmax, gscale);
+  else
+    hipLaunchKernelGGL(k_topk_scatter_var<float>, grid, block, 0, stream, dst,
+                       hdr, idx, (const float*)val, kmax, gscale);
   return (int)hipGetLastError();
 }
 

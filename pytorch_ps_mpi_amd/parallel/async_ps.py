@@ -162,8 +162,7 @@ class AsyncPSEngine:
         self.wseg = []
         off = 0
         for b in flat.buckets:
-            wn = codec.wire_numel(b.numel, flat.dtype) \
-                if codec.name == "topk" else codec.wire_numel(b.numel)
+            wn = codec.wire_numel(b.numel, flat.dtype)
             self.wseg.append((off, wn))
             off += (wn + WSEG_ALIGN - 1) // WSEG_ALIGN * WSEG_ALIGN
         self.wire_total = off

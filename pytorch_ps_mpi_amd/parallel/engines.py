@@ -76,8 +76,7 @@ class ReplicatedEngine:
             self.wire_send = {}
             self.wire_slots = {}
             for b in flat.buckets:
-                wn = codec.wire_numel(b.numel, flat.dtype) \
-                    if codec.name == "topk" else codec.wire_numel(b.numel)
+                wn = codec.wire_numel(b.numel, flat.dtype)
                 wd = codec.wire_dtype(flat.dtype)
                 self.wire_send[b.idx] = torch.zeros(wn, dtype=wd, device=dev)
                 self.wire_slots[b.idx] = [
@@ -176,8 +175,7 @@ class SyncPSEngine:
             if ident:
                 self.wire_send[b.idx] = None  # grad view used directly
             else:
-                wn = codec.wire_numel(b.numel, flat.dtype) \
-                    if codec.name == "topk" else codec.wire_numel(b.numel)
+                wn = codec.wire_numel(b.numel, flat.dtype)
                 wd = codec.wire_dtype(flat.dtype)
                 self.wire_send[b.idx] = torch.zeros(wn, dtype=wd, device=dev)
             if comm.is_ps:

@@ -176,3 +176,34 @@ def test_async_staleness_from_push_time(tmp_path):
     out = str(tmp_path / "ok.txt")
     mp.spawn(_stale_worker, args=(_free_port(), out), nprocs=2, join=True)
     assert os.path.exists(out)
+
+
+def _topkt_worker(rank, port, out_file):
+    from pytorch_ps_mpi_amd import SGD, models
+    _setup(rank, 2, port)
+    torch.manual_seed(0)
+    model = models.build_model("mlp")
+    x, y = models.synthetic_batch("mlp", 16, seed=rank + 1)
+    opt = SGD(model.named_parameters(), lr=0.05, momentum=0.9, mode="async",
+              code="topkt:0.02:0.3", bucket_mb=0.05, window=2, max_stale=4)
+    losses = []
+    for _ in range(10):
+        opt.zero_grad()
+        loss = models.loss_fn("mlp", model, x, y)
+        loss.backward()
+        l, m = opt.step(loss=loss)
+        losses.append(float(l.detach()))
+    opt.finish()
+    assert all(torch.isfinite(torch.tensor(losses)))
+    assert losses[-1] < losses[0]
+    if rank == 0:
+        with open(out_file, "w") as f:
+            f.write("ok")
+
+
+def test_async_topk_threshold_codec(tmp_path):
+    """Variable-k wire over the async protocol (fixed capacity, device-side
+    used-length header)."""
+    out = str(tmp_path / "ok.txt")
+    mp.spawn(_topkt_worker, args=(_free_port(), out), nprocs=2, join=True)
+    assert os.path.exists(out)

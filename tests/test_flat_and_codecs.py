@@ -256,3 +256,44 @@ def test_host_codec_explicit_capacity():
 
     c = codecs.HostCodec(Dense(), capacity=1 << 20)
     assert c.wire_numel(123) == 1 << 20
+
+
+def test_topk_threshold_codec_variable_k():
+    """Device-side variable-length wire: k_used is data-dependent and rides
+    in the wire header; decode touches exactly the used span."""
+    c = codecs.TopKThreshold(alpha=0.25, max_density=0.5)
+    n = 4096
+    torch.manual_seed(3)
+    # two magnitude populations: 32 spikes far above the rest
+    src = torch.randn(n) * 0.01
+    spikes = torch.randperm(n)[:32]
+    src[spikes] = torch.sign(torch.randn(32)) * (1.0 + torch.rand(32))
+    wn = c.wire_numel(n, torch.float32)
+    wire = torch.zeros(wn, dtype=torch.uint8)
+    c.encode(src, wire)
+    k, hdr, idx, val = c._views(wire, n, torch.float32)
+    k_used = int(hdr[0])
+    assert 32 <= k_used < k, f"k_used {k_used} should be data-dependent"
+    dst = torch.zeros(n)
+    c.decode_reduce(dst, [wire], src_dtype=torch.float32)
+    nz = dst.nonzero().flatten()
+    assert len(nz) == k_used
+    assert torch.allclose(dst[nz], src[nz])
+    # every spike is included (they are within alpha of the peak)
+    assert torch.allclose(dst[spikes], src[spikes])
+
+
+def test_topk_threshold_kmax_cap():
+    c = codecs.TopKThreshold(alpha=1e-6, max_density=0.01)  # selects all -> cap
+    n = 10000
+    src = torch.randn(n)
+    wire = torch.zeros(c.wire_numel(n, torch.float32), dtype=torch.uint8)
+    c.encode(src, wire)
+    k, hdr, _, _ = c._views(wire, n, torch.float32)
+    assert int(hdr[0]) == k == c.kmax_for(n)
+
+
+def test_get_codec_topkt_spec():
+    c = codecs.get_codec("topkt:0.1:0.2")
+    assert isinstance(c, codecs.TopKThreshold)
+    assert abs(c.alpha - 0.1) < 1e-9 and abs(c.max_density - 0.2) < 1e-9

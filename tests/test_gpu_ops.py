@@ -186,3 +186,54 @@ def test_topk_scatter_gpu():
     ref = torch.zeros(n)
     ref[idx.cpu().long()] = 2.0 * val.cpu()
     assert torch.allclose(dst.cpu(), ref)
+
+
+def test_topk_thresh_gpu_invariants():
+    """Variable-k threshold select: k_used is content-dependent, lands in
+    the device header, and the selected set obeys the key-bin threshold."""
+    torch.manual_seed(11)
+    n = 200_000
+    kmax = 20_000
+    x = (torch.randn(n) * 0.01)
+    spikes = torch.randperm(n)[:500]
+    x[spikes] = torch.sign(torch.randn(500)) * (1.0 + torch.rand(500))
+    x = x.bfloat16().to(_dev())
+    ws = ops.topk_workspace(_dev())
+    hdr = torch.zeros(1, dtype=torch.int32, device=_dev())
+    idx = torch.zeros(kmax, dtype=torch.int32, device=_dev())
+    val = torch.zeros(kmax, dtype=torch.bfloat16, device=_dev())
+    off = 16  # alpha = 2^-2 = 0.25
+    ops.topk_thresh_encode(x, off, kmax, ws, hdr, idx, val)
+    k = int(hdr.item())
+    assert 500 <= k < kmax, k
+    idx_c = idx[:k].cpu().long()
+    assert len(set(idx_c.tolist())) == k
+    x_c = x.float().cpu()
+    assert torch.equal(val[:k].float().cpu(), x_c[idx_c])
+    keys = (x_c.abs().view(torch.int32) >> 21) & 0x7FF
+    sel = torch.zeros(n, dtype=torch.bool)
+    sel[idx_c] = True
+    assert keys[sel].min() >= keys[~sel].max()
+    # k matches the key-threshold census (cap not hit here)
+    thr0 = max(0, int(keys.max()) - off)
+    assert k == min(int((keys >= thr0).sum()), kmax)
+    # scatter_var reads exactly k entries
+    dst = torch.zeros(n, dtype=torch.float32, device=_dev())
+    ops.topk_scatter_var(dst, hdr, idx, val, kmax, gscale=2.0)
+    dc = dst.cpu()
+    nz = dc.nonzero().flatten()
+    assert len(nz) == k
+    assert torch.allclose(dc[nz], 2.0 * x_c[nz])
+
+
+def test_topk_thresh_gpu_kmax_cap():
+    torch.manual_seed(12)
+    n = 50_000
+    kmax = 100
+    x = torch.randn(n, device=_dev())  # smooth -> census >> kmax
+    ws = ops.topk_workspace(_dev())
+    hdr = torch.zeros(1, dtype=torch.int32, device=_dev())
+    idx = torch.zeros(kmax, dtype=torch.int32, device=_dev())
+    val = torch.zeros(kmax, dtype=torch.float32, device=_dev())
+    ops.topk_thresh_encode(x, 80, kmax, ws, hdr, idx, val)
+    assert int(hdr.item()) == kmax
