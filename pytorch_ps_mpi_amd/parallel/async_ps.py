@@ -81,9 +81,28 @@ WSEG_ALIGN = 256  # wire-segment alignment (elements) — keeps every bucket
 #                   segment 16B-vectorizable and int32-viewable for the codecs
 
 
+_poll_streams = {}
+
+
+def _tag_value(t):
+    """Read an arrival tag.  On GPU the D2H copy goes on a dedicated side
+    stream: .item() on the current stream would queue behind every
+    outstanding compute kernel — a full stream drain per poll.  Safe
+    because tags are written by the comm stream and polled monotonically
+    (0 -> nonzero)."""
+    if t.is_cuda:
+        s = _poll_streams.get(t.device)
+        if s is None:
+            s = torch.cuda.Stream(device=t.device)
+            _poll_streams[t.device] = s
+        with torch.cuda.stream(s):
+            return int(t.item())
+    return int(t.item())
+
+
 def _tagged(slot):
     """True when the slot's tail tag landed => the whole message landed."""
-    return slot["reqs"] is not None and int(slot["tail"].item()) != 0
+    return slot["reqs"] is not None and _tag_value(slot["tail"]) != 0
 
 
 def _wait(reqs, timeout_s=None, what=""):
