@@ -561,11 +561,37 @@ class AsyncPSEngine:
         self._apply_fn = apply_fn
         comm = self.comm
         if comm.world <= 1:
-            flat = self.flat
-            for b in flat.buckets:
-                ops.reduce_accum(flat.agg_view(b), [flat.grad_view(b)],
-                                 scale=self.gscale, beta=0.0)
-                apply_fn(b)
+            flat, codec = self.flat, self.codec
+            if codec.name == "identity":
+                for b in flat.buckets:
+                    with metrics.timer("decode_time"):
+                        ops.reduce_accum(flat.agg_view(b),
+                                         [flat.grad_view(b)],
+                                         scale=self.gscale, beta=0.0)
+                    with metrics.timer("optim_step_time"):
+                        apply_fn(b)
+            else:
+                # lossy codec at world 1: run the REAL encode->decode wire
+                # round trip per bucket (what every worker pays and what the
+                # PS applies), so a 1-GPU bench point of a codec config has
+                # the codec's cost and its numerics
+                if not hasattr(self, "_loop_wire"):
+                    self._loop_wire = torch.zeros(
+                        self.wire_total, dtype=self.wire_dtype,
+                        device=flat.flat_param.device)
+                for b, (off, wn) in zip(flat.buckets, self.wseg):
+                    seg = self._loop_wire[off:off + wn]
+                    with metrics.timer("code_wait"):
+                        codec.encode(flat.grad_view(b), seg)
+                    with metrics.timer("decode_time"):
+                        codec.decode_reduce(flat.agg_view(b), [seg],
+                                            gscale=self.gscale, beta=0.0,
+                                            src_dtype=flat.dtype)
+                    with metrics.timer("optim_step_time"):
+                        apply_fn(b)
+                metrics.add("packaged_bytes",
+                            sum(wn for _, wn in self.wseg)
+                            * self.wire_dtype.itemsize)
             self.ps_version += 1
             return
         if comm.is_ps:
