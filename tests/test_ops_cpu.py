@@ -103,3 +103,32 @@ def test_topk_cpu():
         assert abs(dst[i] - x[i]) < 1e-6
     assert dst.abs().sum() > 0
     assert (dst[[i for i in range(n) if i not in sel]] == 0).all()
+
+
+def test_reduce_accum_many_sources():
+    """No 8-source cliff: 12 sources chunk into kernel passes (world > 8)."""
+    torch.manual_seed(0)
+    n = 1024
+    srcs = [torch.randn(n) for _ in range(12)]
+    dst = torch.full((n,), 2.0)
+    ops.reduce_accum(dst, srcs, scale=0.5, beta=0.25)
+    ref = 0.25 * torch.full((n,), 2.0) + 0.5 * sum(srcs)
+    assert torch.allclose(dst, ref, atol=1e-5)
+
+
+def test_quant8_reduce_many_sources():
+    torch.manual_seed(1)
+    n = 700
+    nc = ops.quant8_nscales(n)
+    scales, qs, ref = [], [], torch.zeros(n)
+    for _ in range(11):
+        src = torch.randn(n)
+        s = torch.zeros(nc)
+        q = torch.zeros(n, dtype=torch.int8)
+        ops.quant8_encode(src, s, q)
+        scales.append(s)
+        qs.append(q)
+        ref += s.repeat_interleave(256)[:n] * q.float()
+    dst = torch.zeros(n)
+    ops.quant8_reduce(dst, scales, qs, gscale=2.0)
+    assert torch.allclose(dst, 2.0 * ref, atol=1e-4)
