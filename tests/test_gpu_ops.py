@@ -237,3 +237,13 @@ def test_topk_thresh_gpu_kmax_cap():
     val = torch.zeros(kmax, dtype=torch.float32, device=_dev())
     ops.topk_thresh_encode(x, 80, kmax, ws, hdr, idx, val)
     assert int(hdr.item()) == kmax
+
+
+def test_reduce_accum_many_sources_gpu():
+    torch.manual_seed(0)
+    n = 8192
+    srcs = [torch.randn(n).bfloat16().to(_dev()) for _ in range(12)]
+    dst = torch.full((n,), 2.0, device=_dev())
+    ops.reduce_accum(dst, srcs, scale=0.5, beta=0.25)
+    ref = 0.25 * torch.full((n,), 2.0) + 0.5 * sum(s.float().cpu() for s in srcs)
+    assert torch.allclose(dst.cpu(), ref, atol=1e-2)
