@@ -1,8 +1,9 @@
 // attn_kernels.hip — flash-attention fwd+bwd for CDNA4 (gfx950).
 //
-// Status: correctness-first; hardware-validated against torch SDPA
-// (tests/test_gpu_attn.py) but not yet perf-competitive with aotriton —
-// the models keep torch SDPA until the MFMA-tiled rewrite (ROADMAP item 2).
+// Status: the ON-GPU ORACLE for the MFMA-tiled production kernels
+// (mfma_attn_kernels.hip, round 2): exact shape-generic one-wave-per-row
+// flash fwd+bwd, hardware-validated against torch SDPA
+// (tests/test_gpu_attn.py, PS_AMD_ATTN=ref).  Slow by design (~2 TF).
 //
 // Shape contract: q,k,v,o,do,dq,dk,dv are [B, H, N, D] bf16 contiguous with
 // D == 64 (one lane per head dim).  Online-softmax forward saves

@@ -1,0 +1,19 @@
+#!/usr/bin/env python3
+"""GPT-2-small training-step driver for rocprofv3 (profiles/gpt2_steady)."""
+import os, sys, time
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import torch
+from pytorch_ps_mpi_amd import Adam, models
+
+dev = torch.device("cuda:0")
+torch.manual_seed(0)
+model = models.build_model("gpt2_small", device=dev, dtype=torch.bfloat16)
+opt = Adam(model.named_parameters(), lr=1e-4, mode="async", grad_scale="mean")
+x, y = models.synthetic_batch("gpt2_small", 96, device=dev,
+                              dtype=torch.bfloat16, seed=1, seq_len=512)
+for _ in range(6):
+    opt.zero_grad()
+    models.loss_fn("gpt2_small", model, x, y).backward()
+    opt.step()
+torch.cuda.synchronize()
+print("done")
