@@ -8,7 +8,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..ops.attn import fused_sdpa
+from ..ops.attn import fused_sdpa_qkv
 from ..ops.ce import fused_cross_entropy
 from ..ops.ln import FusedLayerNorm
 
@@ -27,8 +27,7 @@ class CausalBlock(nn.Module):
     def forward(self, x):
         B, T, D = x.shape
         qkv = self.qkv(self.n1(x)).view(B, T, 3, self.heads, D // self.heads)
-        q, k, v = qkv.permute(2, 0, 3, 1, 4)
-        y = fused_sdpa(q, k, v, is_causal=True)
+        y = fused_sdpa_qkv(qkv, is_causal=True)  # [B, H, T, Dh]
         y = y.transpose(1, 2).reshape(B, T, D)
         x = x + self.proj(y)
         x = x + self.fc2(F.gelu(self.fc1(self.n2(x))))

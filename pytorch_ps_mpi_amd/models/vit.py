@@ -6,7 +6,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..ops.attn import fused_sdpa
+from ..ops.attn import fused_sdpa_qkv
 from ..ops.ln import FusedLayerNorm
 
 
@@ -25,8 +25,7 @@ class Block(nn.Module):
     def forward(self, x):
         B, N, D = x.shape
         qkv = self.qkv(self.n1(x)).view(B, N, 3, self.heads, D // self.heads)
-        q, k, v = qkv.permute(2, 0, 3, 1, 4)
-        y = fused_sdpa(q, k, v)
+        y = fused_sdpa_qkv(qkv)  # [B, H, N, Dh]
         y = y.transpose(1, 2).reshape(B, N, D)
         x = x + self.proj(y)
         x = x + self.fc2(F.gelu(self.fc1(self.n2(x))))

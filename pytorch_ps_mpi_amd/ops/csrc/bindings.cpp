@@ -88,7 +88,7 @@ int ps_fa_bwd(void* stream, const void* q, const void* k, const void* v,
               const void* o, const void* dout, const float* lse, float* delta,
               void* dq, void* dk, void* dv, int64_t BH, int64_t N,
               float scale, int causal, int64_t H, int64_t sB, int64_t sH,
-              int64_t sN);
+              int64_t sN, int64_t oB, int64_t oH, int64_t oN);
 int ps_fa_selfcheck(void* stream, const void* a, const void* b, float* c);
 }
 
@@ -549,16 +549,19 @@ void fa_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o,
             at::Tensor dk, at::Tensor dv, int64_t N, double scale,
             bool causal) {
   fa_check_qkv(q, k, v);
-  TORCH_CHECK(o.is_contiguous() && dout.is_contiguous() &&
-                  dq.is_contiguous() && dk.is_contiguous() &&
-                  dv.is_contiguous(),
-              "o/dout/dq/dk/dv must be contiguous");
+  // dq/dk/dv may be strided views of ONE dqkv buffer (identical strides,
+  // unit d-stride, 16B-aligned rows) -> backward feeds the fused qkv grad
+  // directly and autograd's CatArrayBatchedCopy pass disappears
+  fa_check_qkv(dq, dk, dv);
+  TORCH_CHECK(o.is_contiguous() && dout.is_contiguous(),
+              "o/dout must be contiguous");
   const int64_t BH = q.size(0) * q.size(1);
   throw_on(ps_fa_bwd(cur_stream(q), q.data_ptr(), k.data_ptr(), v.data_ptr(),
                      o.data_ptr(), dout.data_ptr(), lse.data_ptr<float>(),
                      delta.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
                      dv.data_ptr(), BH, N, (float)scale, causal ? 1 : 0,
-                     q.size(1), q.stride(0), q.stride(1), q.stride(2)),
+                     q.size(1), q.stride(0), q.stride(1), q.stride(2),
+                     dq.stride(0), dq.stride(1), dq.stride(2)),
            "fa_bwd");
 }
 

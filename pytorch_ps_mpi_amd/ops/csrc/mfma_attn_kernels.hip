@@ -311,7 +311,8 @@ k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
             const __hip_bfloat16* __restrict__ dout,
             const float* __restrict__ lse, const float* __restrict__ delta,
             __hip_bfloat16* __restrict__ dq, int64_t N, float scale,
-            int causal, int64_t H, int64_t sB, int64_t sH, int64_t sN) {
+            int causal, int64_t H, int64_t sB, int64_t sH, int64_t sN,
+            int64_t oB, int64_t oH, int64_t oN) {
   // lds: Kt double buffer [2][64][32] + per-wave dS [8][32][32]
   __shared__ __align__(16) __hip_bfloat16 lds[2 * FA_D * FA_BNB
                                               + 8 * 32 * FA_BNB];
@@ -427,7 +428,7 @@ k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
     }
   }
 
-  __hip_bfloat16* dqb = dq + bh * N * FA_D;
+  __hip_bfloat16* dqb = dq + (bh / H) * oB + (bh % H) * oH;
 #pragma unroll
   for (int rf = 0; rf < 2; ++rf)
 #pragma unroll
@@ -436,7 +437,7 @@ k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
       if (row >= N) continue;
 #pragma unroll
       for (int df = 0; df < 4; ++df)
-        dqb[row * FA_D + df * 16 + c] =
+        dqb[row * oN + df * 16 + c] =
             (__hip_bfloat16)(acc[rf][df][r] * scale);
     }
 }
@@ -449,7 +450,8 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
              const float* __restrict__ lse, const float* __restrict__ delta,
              __hip_bfloat16* __restrict__ dk, __hip_bfloat16* __restrict__ dv,
              int64_t N, float scale, int causal,
-             int64_t H, int64_t sB, int64_t sH, int64_t sN) {
+             int64_t H, int64_t sB, int64_t sH, int64_t sN,
+             int64_t oB, int64_t oH, int64_t oN) {
   // lds: double-buffered {Qt,dOt} [2][2][64][32] + per-wave P/dS [8][32][32]
   __shared__ __align__(16) __hip_bfloat16 lds[4 * FA_D * FA_BNB
                                               + 8 * 32 * FA_BNB];
@@ -590,8 +592,8 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
     }
   }
 
-  __hip_bfloat16* dkb = dk + bh * N * FA_D;
-  __hip_bfloat16* dvb = dv + bh * N * FA_D;
+  __hip_bfloat16* dkb = dk + (bh / H) * oB + (bh % H) * oH;
+  __hip_bfloat16* dvb = dv + (bh / H) * oB + (bh % H) * oH;
 #pragma unroll
   for (int rf = 0; rf < 2; ++rf)
 #pragma unroll
@@ -600,9 +602,9 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
       if (row >= N) continue;
 #pragma unroll
       for (int df = 0; df < 4; ++df) {
-        dkb[row * FA_D + df * 16 + c] =
+        dkb[row * oN + df * 16 + c] =
             (__hip_bfloat16)(dK[rf][df][r] * scale);
-        dvb[row * FA_D + df * 16 + c] = (__hip_bfloat16)dV[rf][df][r];
+        dvb[row * oN + df * 16 + c] = (__hip_bfloat16)dV[rf][df][r];
       }
     }
 }
@@ -654,7 +656,7 @@ int ps_fa_bwd(void* stream_, const void* q, const void* k, const void* v,
               const void* o, const void* dout, const float* lse, float* delta,
               void* dq, void* dk, void* dv, int64_t BH, int64_t N,
               float scale, int causal, int64_t H, int64_t sB, int64_t sH,
-              int64_t sN) {
+              int64_t sN, int64_t oB, int64_t oH, int64_t oN) {
   hipStream_t s = (hipStream_t)stream_;
   const int64_t rows = BH * N;
   int64_t db = (rows + 3) / 4;
@@ -667,12 +669,12 @@ int ps_fa_bwd(void* stream_, const void* q, const void* k, const void* v,
                      (const __hip_bfloat16*)q, (const __hip_bfloat16*)k,
                      (const __hip_bfloat16*)v, (const __hip_bfloat16*)dout,
                      lse, delta, (__hip_bfloat16*)dq, N, scale, causal,
-                     H, sB, sH, sN);
+                     H, sB, sH, sN, oB, oH, oN);
   hipLaunchKernelGGL(k_fa_bwd_dkv, grid, dim3(512), 0, s,
                      (const __hip_bfloat16*)q, (const __hip_bfloat16*)k,
                      (const __hip_bfloat16*)v, (const __hip_bfloat16*)dout,
                      lse, delta, (__hip_bfloat16*)dk, (__hip_bfloat16*)dv, N,
-                     scale, causal, H, sB, sH, sN);
+                     scale, causal, H, sB, sH, sN, oB, oH, oN);
   return (int)hipGetLastError();
 }
 

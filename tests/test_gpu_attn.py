@@ -108,3 +108,27 @@ def test_fused_sdpa_strided_qkv(monkeypatch):
                                          is_causal=True)
     o = fused_sdpa(q, k, v, is_causal=True)
     assert (o.float() - ref).abs().max().item() < 0.05
+
+
+def test_fused_sdpa_qkv_function(monkeypatch):
+    """qkv-level Function: strided slices in, assembled dqkv out — fwd and
+    grads vs the fp32 autograd oracle."""
+    monkeypatch.setenv("PS_AMD_ATTN", "mfma")
+    torch.manual_seed(2)
+    dev = "cuda:0"
+    B, T, H, D = 2, 160, 3, 64
+    from pytorch_ps_mpi_amd.ops.attn import fused_sdpa_qkv
+    qkv32 = torch.randn(B, T, 3, H, D, device=dev).bfloat16().float()
+    qkv32.requires_grad_(True)
+    q, k, v = qkv32.permute(2, 0, 3, 1, 4)
+    ref = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+    g = torch.randn_like(ref).bfloat16().float()
+    ref.backward(g)
+
+    qkv = qkv32.detach().bfloat16().requires_grad_(True)
+    o = fused_sdpa_qkv(qkv, is_causal=True)
+    o.backward(g.bfloat16())
+    assert (o.float() - ref).abs().max().item() < 0.05
+    err = (qkv.grad.float() - qkv32.grad).abs().max().item()
+    scale = qkv32.grad.abs().max().item() + 1.0
+    assert err < 0.05 * scale, f"dqkv err {err}"
