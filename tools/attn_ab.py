@@ -24,28 +24,15 @@ def bench(model_name, kw, steps=8, warmup=3):
     torch.cuda.synchronize()
     return (time.perf_counter() - t0) / steps * 1e3
 
-import pytorch_ps_mpi_amd.ops.attn as A
-orig = A.fused_sdpa
 res = {}
 for rnd in range(2):
-    for impl, contig in [("torch", 0), ("mfma", 0), ("mfma", 1)]:
+    for impl in ("torch", "mfma"):
         os.environ["PS_AMD_ATTN"] = impl
-        if contig:
-            def wrap(q, k, v, is_causal=False):
-                return orig(q.contiguous(), k.contiguous(), v.contiguous(),
-                            is_causal=is_causal)
-            A_fn = wrap
-        else:
-            A_fn = orig
-        import pytorch_ps_mpi_amd.models.gpt2 as G
-        import pytorch_ps_mpi_amd.models.vit as V
-        G.fused_sdpa = A_fn
-        V.fused_sdpa = A_fn
         for mn, kw in [("gpt2_small", {"batch": 96}), ("vit_b16", {"batch": 512})]:
             ms = bench(mn, kw)
-            key = (mn, impl, contig)
+            key = (mn, impl)
             res.setdefault(key, []).append(ms)
-            print(f"round{rnd} {mn:10s} {impl}{'+contig' if contig else ''}: {ms:7.2f} ms", flush=True)
+            print(f"round{rnd} {mn:10s} {impl}: {ms:7.2f} ms", flush=True)
 print("== medians ==")
 for k, v in sorted(res.items()):
     print(k, round(sorted(v)[len(v)//2], 2))
