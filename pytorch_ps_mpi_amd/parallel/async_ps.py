@@ -101,8 +101,16 @@ def _tag_value(t):
 
 
 def _tagged(slot):
-    """True when the slot's tail tag landed => the whole message landed."""
-    return slot["reqs"] is not None and _tag_value(slot["tail"]) != 0
+    """True when the slot's tail tag landed => the whole message landed.
+
+    The check is an EXACT match against the slot's expected sequence value
+    (stop markers are negative): a reused slot's previous tag, or a re-arm
+    zero_() still in flight on another stream, can never read as "arrived",
+    so a non-blocking poll never turns into a blocking wait."""
+    if slot["reqs"] is None:
+        return False
+    v = _tag_value(slot["tail"])
+    return v == slot["expect"] or v < 0
 
 
 def _wait(reqs, timeout_s=None, what=""):
@@ -127,7 +135,8 @@ class _Peer:
     """PS-side per-worker state: recv ring + reply ring + reply-shard cursor."""
 
     __slots__ = ("rank", "push_g", "reply_g", "slots", "head", "replies",
-                 "rhead", "cursor", "stopped", "dropped", "last_seen")
+                 "rhead", "cursor", "stopped", "dropped", "last_seen",
+                 "post_seq")
 
     def __init__(self, rank, push_g, reply_g, ring, reply_ring, wire_total,
                  wdt, shard_max, pdt, dev):
@@ -139,7 +148,11 @@ class _Peer:
             "wire": torch.zeros(wire_total, dtype=wdt, device=dev),
             "tail": torch.zeros(1, dtype=torch.int64, device=dev),
             "reqs": None,
+            "expect": 0,
         } for _ in range(ring)]
+        self.post_seq = 0  # worker pushes are numbered 1,2,...; a slot's
+        #                    expected tail = the sequence of the push that
+        #                    will land in it (stop markers are -1)
         self.head = 0
         self.replies = [{
             "hdr": torch.zeros(2, dtype=torch.int64, device=dev),
@@ -241,6 +254,7 @@ class AsyncPSEngine:
                 "reqs": None,
                 "push_step": 0,
                 "shard": 0,
+                "expect": 0,
             } for _ in range(self.window)]
             self.sent = 0
             self.harvested = 0
@@ -301,6 +315,8 @@ class AsyncPSEngine:
                 self._post(st, s)
 
     def _post(self, st, slot):
+        st.post_seq += 1
+        slot["expect"] = st.post_seq
         slot["tail"].zero_()  # re-arm the arrival tag BEFORE posting
         reqs = [dist.irecv(slot["hdr"], src=st.rank, group=st.push_g)]
         for off, wn in self.wseg:
@@ -524,6 +540,7 @@ class AsyncPSEngine:
         lo, hi = self.shards[sh]
         rsl["shard"] = sh
         rsl["push_step"] = psh["step"]
+        rsl["expect"] = self.cursor  # PS reply tail = its cursor (mirrored)
         rsl["tail"].zero_()
         psh["tail"].fill_(self.worker_step)
         with metrics.timer("isend_time"):
