@@ -168,6 +168,7 @@ k_fa_fwd(const __hip_bfloat16* __restrict__ q,
       fa_fill_t(vb, (t + 1) * FA_BN, N,
                 lds + ((t + 1) & 1) * FA_D * FA_BN, sN);
     const __hip_bfloat16* vt = lds + (t & 1) * FA_D * FA_BN;
+    if (row0 >= N) continue;                 // whole wave past the rows
     if (causal && k0 > row0 + 31) continue;  // above this wave's diagonal
 
     // S = Q @ K^T  (B-frag: 8 consecutive d of key row -> direct load).
@@ -362,6 +363,7 @@ k_fa_bwd_dq(const __hip_bfloat16* __restrict__ q,
       fa_fill_t32(kb, (t + 1) * FA_BNB,
                   N, lds + ((t + 1) & 1) * FA_D * FA_BNB, sN);
     const __hip_bfloat16* kt = lds + (t & 1) * FA_D * FA_BNB;
+    if (row0 >= N) continue;  // whole wave past the rows
     if (causal && k0 > row0 + 31) continue;
 
     // S and dP in one pass over kk (batched B-frag loads: latencies overlap)
@@ -501,6 +503,7 @@ k_fa_bwd_dkv(const __hip_bfloat16* __restrict__ q,
     }
     const __hip_bfloat16* qt = lds + (t & 1) * 2 * FA_D * FA_BNB;
     const __hip_bfloat16* dot = qt + FA_D * FA_BNB;
+    if (key0 >= N) continue;  // whole wave past the keys
     if (causal && q0 + FA_BNB - 1 < key0) continue;  // below diagonal
 
     float lse_c[2], dl_c[2];
