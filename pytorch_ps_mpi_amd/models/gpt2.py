@@ -10,6 +10,7 @@ import torch.nn.functional as F
 
 from ..ops.attn import fused_sdpa_qkv
 from ..ops.ce import fused_cross_entropy
+from ..ops.linear import FusedLinear
 from ..ops.ln import FusedLayerNorm
 
 
@@ -17,11 +18,11 @@ class CausalBlock(nn.Module):
     def __init__(self, dim, heads):
         super().__init__()
         self.n1 = FusedLayerNorm(dim)
-        self.qkv = nn.Linear(dim, 3 * dim)
-        self.proj = nn.Linear(dim, dim)
+        self.qkv = FusedLinear(dim, 3 * dim)
+        self.proj = FusedLinear(dim, dim)
         self.n2 = FusedLayerNorm(dim)
-        self.fc1 = nn.Linear(dim, 4 * dim)
-        self.fc2 = nn.Linear(4 * dim, dim)
+        self.fc1 = FusedLinear(dim, 4 * dim)
+        self.fc2 = FusedLinear(4 * dim, dim)
         self.heads = heads
 
     def forward(self, x):

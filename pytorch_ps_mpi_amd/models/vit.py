@@ -7,6 +7,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..ops.attn import fused_sdpa_qkv
+from ..ops.linear import FusedLinear
 from ..ops.ln import FusedLayerNorm
 
 
@@ -14,13 +15,13 @@ class Block(nn.Module):
     def __init__(self, dim, heads, mlp_ratio=4.0):
         super().__init__()
         self.n1 = FusedLayerNorm(dim)
-        self.qkv = nn.Linear(dim, dim * 3)
-        self.proj = nn.Linear(dim, dim)
+        self.qkv = FusedLinear(dim, dim * 3)
+        self.proj = FusedLinear(dim, dim)
         self.heads = heads
         self.n2 = FusedLayerNorm(dim)
         h = int(dim * mlp_ratio)
-        self.fc1 = nn.Linear(dim, h)
-        self.fc2 = nn.Linear(h, dim)
+        self.fc1 = FusedLinear(dim, h)
+        self.fc2 = FusedLinear(h, dim)
 
     def forward(self, x):
         B, N, D = x.shape

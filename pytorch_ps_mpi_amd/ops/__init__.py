@@ -242,4 +242,5 @@ def topk_scatter_var(dst, hdr, idx, val, kmax, gscale=1.0):
 from . import bn  # noqa: E402,F401  (fused BatchNorm module; needs ops ready)
 from . import ln  # noqa: E402,F401
 from . import ce  # noqa: E402,F401
-from . import attn  # noqa: E402,F401  (experimental)
+from . import attn  # noqa: E402,F401
+from . import linear  # noqa: E402,F401

@@ -69,6 +69,8 @@ int ps_ln_bwd_dx(void* stream, const void* x, const void* dy, void* dx,
 int ps_ln_bwd_dgb(void* stream, const void* x, const void* dy,
                   const float* mean, const float* rstd, float* dgamma,
                   float* dbeta, int64_t rows, int64_t D);
+int ps_colsum(void* stream, const void* dy, float* out, int64_t rows,
+              int64_t D);
 int ps_ce_fwd(void* stream, const void* logits, const int64_t* targets,
               float* losses, float* lse, int64_t T, int64_t V);
 int ps_ce_bwd(void* stream, const void* logits, const int64_t* targets,
@@ -252,6 +254,17 @@ void quant8_reduce(at::Tensor dst, std::vector<at::Tensor> scales,
                             (int)qs.size(), dst.numel(), (float)gscale,
                             (float)beta),
            "quant8_reduce");
+}
+
+void colsum(at::Tensor dy, at::Tensor out) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && dy.dim() == 2 &&
+                  dy.scalar_type() == at::kBFloat16,
+              "dy must be contiguous 2D bf16");
+  check_flat(out, at::kFloat, "out");
+  TORCH_CHECK(out.numel() == dy.size(1), "out length mismatch");
+  throw_on(ps_colsum(cur_stream(dy), dy.data_ptr(), out.data_ptr<float>(),
+                     dy.size(0), dy.size(1)),
+           "colsum");
 }
 
 int64_t topk_workspace_words() { return ps_topk_workspace_words(); }
@@ -582,6 +595,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fa_selfcheck", &fa_selfcheck, "16x16x32 MFMA fragment-layout check");
   m.def("topk_encode_thresh", &topk_encode_thresh,
         "variable-k magnitude-threshold select (k_used -> device header)");
+  m.def("colsum", &colsum, "column sum (bias grads)");
   m.def("topk_scatter_var", &topk_scatter_var,
         "scatter-add with device-side k header");
   m.def("ce_fwd", &ce_fwd);

@@ -216,7 +216,58 @@ k_ln_bwd_dgb(const __hip_bfloat16* __restrict__ x,
   }
 }
 
+// plain column sum (Linear bias gradients): same shape as k_ln_bwd_dgb
+__global__ void __launch_bounds__(LN_BLOCK)
+k_colsum(const __hip_bfloat16* __restrict__ dy, float* __restrict__ out,
+         int64_t rows, int64_t D) {
+  __shared__ float red[4][64];
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int ch8 = (lane & 7) * 8;
+  const int rsub = lane >> 3;
+  const int64_t cbase = (int64_t)blockIdx.x * 64;
+  float sb[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) sb[j] = 0.0f;
+  const int64_t rstride = (int64_t)gridDim.y * 32;
+  for (int64_t r = (int64_t)blockIdx.y * 32 + wave * 8 + rsub; r < rows;
+       r += rstride) {
+    const bnx8 dv = *(const bnx8*)(dy + r * D + cbase + ch8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) sb[j] += lb2f(dv.v[j]);
+  }
+#pragma unroll
+  for (int d = 8; d < 64; d <<= 1)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) sb[j] += __shfl_xor(sb[j], d, 64);
+  if (lane < 8)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) red[wave][ch8 + j] = sb[j];
+  __syncthreads();
+  if (wave == 0)
+    atomicAdd(&out[cbase + lane],
+              red[0][lane] + red[1][lane] + red[2][lane] + red[3][lane]);
+}
+
 extern "C" {
+
+int ps_colsum(void* stream_, const void* dy, float* out, int64_t rows,
+              int64_t D) {
+  hipStream_t s = (hipStream_t)stream_;
+  if (D % 64 != 0) return 9100;
+  hipError_t e = hipMemsetAsync(out, 0, D * sizeof(float), s);
+  if (e) return (int)e;
+  const int ct = (int)(D / 64);
+  int64_t yb = (rows + 31) / 32;
+  int64_t cap = 1024 / ct;
+  if (cap < 1) cap = 1;
+  if (yb > cap) yb = cap;
+  hipLaunchKernelGGL(k_colsum, dim3((unsigned)ct, (unsigned)yb),
+                     dim3(LN_BLOCK), 0, s, (const __hip_bfloat16*)dy, out,
+                     rows, D);
+  return (int)hipGetLastError();
+}
+
 
 int ps_ln_fwd(void* stream_, const void* x, void* y, const void* gamma,
               const void* beta, float* mean, float* rstd, int64_t rows,

@@ -247,3 +247,31 @@ def test_reduce_accum_many_sources_gpu():
     ops.reduce_accum(dst, srcs, scale=0.5, beta=0.25)
     ref = 0.25 * torch.full((n,), 2.0) + 0.5 * sum(s.float().cpu() for s in srcs)
     assert torch.allclose(dst.cpu(), ref, atol=1e-2)
+
+
+def test_fused_linear_parity():
+    """FusedLinear (custom bias-grad colsum, GEMMs unchanged) matches
+    nn.Linear fwd and all three grads."""
+    import torch.nn as nn
+    from pytorch_ps_mpi_amd.ops.linear import FusedLinear
+    torch.manual_seed(5)
+    dev = _dev()
+    ref = nn.Linear(256, 512).to(dev, torch.bfloat16)
+    fl = FusedLinear(256, 512).to(dev, torch.bfloat16)
+    with torch.no_grad():
+        fl.weight.copy_(ref.weight)
+        fl.bias.copy_(ref.bias)
+    x1 = torch.randn(4, 37, 256, device=dev, dtype=torch.bfloat16,
+                     requires_grad=True)
+    x2 = x1.detach().clone().requires_grad_(True)
+    g = torch.randn(4, 37, 512, device=dev, dtype=torch.bfloat16)
+    y1 = ref(x1)
+    y2 = fl(x2)
+    assert torch.equal(y1, y2)
+    y1.backward(g)
+    y2.backward(g)
+    assert torch.equal(x1.grad, x2.grad)
+    assert torch.allclose(fl.weight.grad.float(), ref.weight.grad.float(),
+                          atol=1e-2, rtol=1e-2)
+    assert torch.allclose(fl.bias.grad.float(), ref.bias.grad.float(),
+                          atol=1e-2, rtol=1e-2)
