@@ -12,6 +12,8 @@ to nn.Linear's own autograd (bitwise the stock semantics).
 
 from __future__ import annotations
 
+import os
+
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
@@ -42,6 +44,7 @@ class FusedLinear(nn.Linear):
     def forward(self, x):
         if (HAVE_EXT and x.is_cuda and x.dtype == torch.bfloat16
                 and self.bias is not None
-                and self.out_features % 64 == 0):
+                and self.out_features % 64 == 0
+                and os.environ.get("PS_AMD_FUSED_LINEAR", "1") != "0"):
             return _LinearFn.apply(x, self.weight, self.bias)
         return super().forward(x)
