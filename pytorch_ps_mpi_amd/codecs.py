@@ -9,7 +9,7 @@ mpi_comms.py:150-163 — pure latency it did not need).
 
 Contract (all tensors device-resident, no host round trips):
     wire_dtype                      -> torch dtype of the wire tensor
-    wire_numel(bucket_numel)        -> fixed wire length for a bucket
+    wire_numel(bucket_numel, src_dtype) -> fixed wire length for a bucket
     encode(src, wire)               -> fill wire from src (model-dtype flat)
     decode_reduce(dst, wires, gscale, beta)
         dst(f32) = beta*dst + gscale * sum_r decode(wires[r])
