@@ -166,3 +166,51 @@ def test_mpi_ps_factory():
     import pytest as _pytest
     with _pytest.raises(ValueError):
         MPI_PS(m.named_parameters(), optim="rmsprop")
+
+
+def test_checkpoint_preserves_fp32_master(tmp_path):
+    """Resume must not round the fp32 master through bf16 (advisor r1):
+    masters whose low mantissa bits differ from their bf16 rounding must
+    survive a save/load cycle bitwise."""
+    import torch.nn as nn
+    from pytorch_ps_mpi_amd import SGD
+    from pytorch_ps_mpi_amd.utils import checkpoint
+    torch.manual_seed(0)
+    model = nn.Linear(32, 32)
+    opt = SGD(model.named_parameters(), lr=0.1, momentum=0.9)
+    # perturb the master below bf16 resolution
+    with torch.no_grad():
+        opt.flat.master.add_(torch.randn_like(opt.flat.master) * 1e-6)
+    ref = opt.flat.master.clone()
+    path = str(tmp_path / "ck.pt")
+    checkpoint.save(path, opt)
+
+    model2 = nn.Linear(32, 32)
+    opt2 = SGD(model2.named_parameters(), lr=0.1, momentum=0.9)
+    checkpoint.load(path, opt2)
+    assert torch.equal(opt2.flat.master, ref), "fp32 master not bit-exact"
+
+
+def test_topkt_multi_message_sum():
+    from pytorch_ps_mpi_amd import codecs
+    c = codecs.TopKThreshold(alpha=0.3, max_density=0.5)
+    n = 2048
+    torch.manual_seed(4)
+    a = torch.randn(n) * 0.01
+    b = torch.randn(n) * 0.01
+    a[:5] = 3.0
+    b[2:8] = -2.0  # different k_used per message
+    wn = c.wire_numel(n, torch.float32)
+    wa, wb = (torch.zeros(wn, dtype=torch.uint8) for _ in range(2))
+    c.encode(a, wa)
+    c.encode(b, wb)
+    ka = int(c._views(wa, n, torch.float32)[1][0])
+    kb = int(c._views(wb, n, torch.float32)[1][0])
+    assert ka != kb, (ka, kb)
+    dst = torch.zeros(n)
+    c.decode_reduce(dst, [wa, wb], gscale=0.5, src_dtype=torch.float32)
+    # indices 2..4 are spikes in BOTH messages: both contributions land
+    assert torch.allclose(dst[2:5], 0.5 * (a[2:5] + b[2:5]), atol=1e-6)
+    # 0..1 only in a, 5..7 only in b
+    assert torch.allclose(dst[:2], 0.5 * a[:2], atol=1e-6)
+    assert torch.allclose(dst[5:8], 0.5 * b[5:8], atol=1e-6)
