@@ -27,22 +27,50 @@ from .. import ops
 
 
 class LocalEngine:
+    """world_size == 1.  A lossy codec still runs its REAL encode->decode
+    round trip per bucket (what one worker would put on the wire and what
+    the PS would apply): a 1-GPU bench point of a codec config carries the
+    codec's cost and numerics."""
+
     name = "local"
 
     def __init__(self, flat, codec, comm, grad_scale=1.0):
         self.flat = flat
         self.codec = codec
         self.gscale = grad_scale
+        self._wire = None
 
     def step(self, apply_fn, metrics):
-        flat = self.flat
+        flat, codec = self.flat, self.codec
+        if codec.name == "identity":
+            for b in flat.buckets:
+                with metrics.timer("decode_time"):
+                    ops.reduce_accum(flat.agg_view(b), [flat.grad_view(b)],
+                                     scale=self.gscale, beta=0.0)
+                with metrics.timer("optim_step_time"):
+                    apply_fn(b)
+            metrics.add("msg_bytes", 0)
+            return
+        if self._wire is None:
+            self._wire = {
+                b.idx: torch.zeros(codec.wire_numel(b.numel, flat.dtype),
+                                   dtype=codec.wire_dtype(flat.dtype),
+                                   device=flat.flat_param.device)
+                for b in flat.buckets
+            }
         for b in flat.buckets:
+            w = self._wire[b.idx]
+            with metrics.timer("code_wait"):
+                codec.encode(flat.grad_view(b), w)
             with metrics.timer("decode_time"):
-                ops.reduce_accum(flat.agg_view(b), [flat.grad_view(b)],
-                                 scale=self.gscale, beta=0.0)
+                codec.decode_reduce(flat.agg_view(b), [w],
+                                    gscale=self.gscale, beta=0.0,
+                                    src_dtype=flat.dtype)
             with metrics.timer("optim_step_time"):
                 apply_fn(b)
-        metrics.add("msg_bytes", 0)
+        metrics.add("packaged_bytes",
+                    sum(w.numel() * w.dtype.itemsize
+                        for w in self._wire.values()))
 
     def finish(self):
         pass

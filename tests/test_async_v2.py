@@ -207,3 +207,35 @@ def test_async_topk_threshold_codec(tmp_path):
     out = str(tmp_path / "ok.txt")
     mp.spawn(_topkt_worker, args=(_free_port(), out), nprocs=2, join=True)
     assert os.path.exists(out)
+
+
+def _accum_worker(rank, port, out_file):
+    """Gradient accumulation (overlap=False) over the async protocol: two
+    backwards per step, codec sees the accumulated gradient."""
+    from pytorch_ps_mpi_amd import SGD, models
+    _setup(rank, 2, port)
+    torch.manual_seed(0)
+    model = models.build_model("mlp")
+    x, y = models.synthetic_batch("mlp", 16, seed=rank + 1)
+    opt = SGD(model.named_parameters(), lr=0.02, momentum=0.9, mode="async",
+              bucket_mb=0.05, window=2, max_stale=4, overlap=False)
+    losses = []
+    for _ in range(6):
+        opt.zero_grad()
+        for h in range(2):  # two micro-batches accumulate
+            loss = models.loss_fn("mlp", model, x[h::2], y[h::2])
+            loss.backward()
+        l, m = opt.step(loss=loss)
+        losses.append(float(l.detach()))
+    opt.finish()
+    assert all(torch.isfinite(torch.tensor(losses)))
+    assert losses[-1] < losses[0]
+    if rank == 0:
+        with open(out_file, "w") as f:
+            f.write("ok")
+
+
+def test_async_grad_accumulation(tmp_path):
+    out = str(tmp_path / "ok.txt")
+    mp.spawn(_accum_worker, args=(_free_port(), out), nprocs=2, join=True)
+    assert os.path.exists(out)
