@@ -187,7 +187,15 @@ class ReplicatedEngine:
 
 
 class SyncPSEngine:
+    """Gather -> PS decode+sum+apply -> broadcast (the reference's README
+    plan).  Per-bucket gathers launch FROM THE BACKWARD HOOKS in fixed
+    bucket order on every rank (gather is a collective, so the schedule must
+    match — same discipline as ReplicatedEngine), overlapping the exchange
+    with the rest of backward; buckets whose hooks never fired launch at
+    step()."""
+
     name = "sync_ps"
+    wants_hooks = True
 
     def __init__(self, flat, codec, comm, grad_scale=1.0):
         self.flat = flat
@@ -199,6 +207,9 @@ class SyncPSEngine:
         self.ident = ident
         self.wire_send = {}
         self.wire_slots = {}
+        self._works = {}
+        self._next = 0
+        self._ready = {}
         for b in flat.buckets:
             if ident:
                 self.wire_send[b.idx] = None  # grad view used directly
@@ -222,22 +233,49 @@ class SyncPSEngine:
             return self.flat.grad_view(b)
         return self.wire_send[b.idx]
 
+    # ---- hook-driven overlap (same schedule rules as ReplicatedEngine) ---
+
+    def start_step(self):
+        self._works = {}
+        self._next = 0
+        self._ready = {b.idx: 0 for b in self.flat.buckets}
+
+    def on_param_grad(self, param):
+        b = self.flat.param_to_bucket.get(param)
+        if b is None or not self._ready:
+            return
+        self._ready[b.idx] += 1
+        if self._ready[b.idx] > len(b.params):
+            raise RuntimeError(
+                "gradient accumulation detected with hook-overlap enabled; "
+                "construct the optimizer with overlap=False to accumulate "
+                "gradients over multiple backward passes")
+        while self._next < len(self.flat.buckets):
+            nb = self.flat.buckets[self._next]
+            if self._ready.get(nb.idx, 0) < len(nb.params):
+                break
+            self._launch(nb)
+            self._next += 1
+
+    def _launch(self, b):
+        if not self.ident:
+            self.codec.encode(self.flat.grad_view(b), self.wire_send[b.idx])
+        gl = self.wire_slots[b.idx] if self.comm.is_ps else None
+        self._works[b.idx] = dist.gather(self._send_wire(b), gather_list=gl,
+                                         dst=self.comm.ps_rank, async_op=True)
+
+    # ---- step ---------------------------------------------------------
+
     def step(self, apply_fn, metrics):
         flat, codec, comm = self.flat, self.codec, self.comm
-        if not self.ident:
-            with metrics.timer("code_wait"):
-                for b in flat.buckets:
-                    codec.encode(flat.grad_view(b), self.wire_send[b.idx])
-        works = []
         with metrics.timer("isend_time"):
-            for b in flat.buckets:
-                gl = self.wire_slots[b.idx] if comm.is_ps else None
-                works.append(dist.gather(self._send_wire(b), gather_list=gl,
-                                         dst=comm.ps_rank, async_op=True))
+            for b in flat.buckets:  # launch whatever the hooks did not
+                if b.idx not in self._works:
+                    self._launch(b)
         bworks = []
-        for b, w in zip(flat.buckets, works):
+        for b in flat.buckets:
             with metrics.timer("comm_wait"):
-                w.wait()
+                self._works[b.idx].wait()
             if comm.is_ps:
                 with metrics.timer("decode_time"):
                     codec.decode_reduce(flat.agg_view(b),
@@ -251,6 +289,8 @@ class SyncPSEngine:
         with metrics.timer("comm_wait"):
             for w in bworks:
                 w.wait()
+        self._works = {}
+        self._ready = {}
         wire_b = flat.total * flat.dtype.itemsize if self.ident else \
             sum(t.numel() * t.dtype.itemsize
                 for t in self.wire_send.values())
